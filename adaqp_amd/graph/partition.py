@@ -1,0 +1,210 @@
+"""Graph partitioning and per-rank LocalGraph construction.
+
+Replaces the reference's DGL/METIS pipeline (``AdaQP/helper/partition.py``,
+``AdaQP/manager/conversion.py``, ``AdaQP/manager/processing.py``) with a
+self-contained implementation:
+
+- assignment: contiguous ``range`` split (the synthetic generator plants
+  locality into ranges, so this matches METIS quality on synthetic data)
+  or greedy ``bfs`` grow (for arbitrary graphs).
+- ``build_local_graph``: reorders one partition into
+  [central | marginal | remote] (parity with ``conversion.py:56-90``),
+  builds the in-edge CSR, global degree slices, and the boundary
+  send/recv structure.
+
+Send/recv order needs NO network handshake (the reference all-gathers
+index maps, ``processing.py:40-79``): both sides sort boundary nodes by
+GLOBAL id, so owner send order == consumer remote order by construction.
+
+Artifact layout parity (reference ``helper/partition.py:41,61-72``,
+``processing.py:76-78``): ``<dir>/<ds>/<P>part/part{i}.pt`` + meta json
+and global degrees — see save_partitions/load_partition.
+"""
+from __future__ import annotations
+
+import json
+import os
+from typing import Dict, List, Optional, Tuple
+
+import torch
+from torch import Tensor
+
+from .csr import LocalGraph, coo_to_csr
+from .synthetic import GlobalGraph
+
+
+# --------------------------------------------------------------------------
+# assignment
+# --------------------------------------------------------------------------
+
+def range_assignment(num_nodes: int, num_parts: int) -> Tensor:
+    bounds = torch.linspace(0, num_nodes, num_parts + 1, dtype=torch.int64)
+    assign = torch.empty(num_nodes, dtype=torch.int64)
+    for p in range(num_parts):
+        assign[bounds[p]:bounds[p + 1]] = p
+    return assign
+
+
+def bfs_assignment(g: GlobalGraph, num_parts: int, seed: int = 0) -> Tensor:
+    """Greedy BFS-grow partitioner: grow each part to ~N/P nodes from a
+    random seed, frontier-first (keeps parts connected -> low edge cut)."""
+    n = g.num_nodes
+    indptr, indices = coo_to_csr(g.dst.clone(), g.src.clone(), n)
+    assign = torch.full((n,), -1, dtype=torch.int64)
+    target = (n + num_parts - 1) // num_parts
+    gen = torch.Generator().manual_seed(seed)
+    perm = torch.randperm(n, generator=gen)
+    cursor = 0
+    for p in range(num_parts):
+        remaining = num_parts - p
+        budget = min(target, n - int((assign >= 0).sum())) if remaining > 1 \
+            else n - int((assign >= 0).sum())
+        count = 0
+        frontier: List[int] = []
+        while count < budget:
+            if not frontier:
+                while cursor < n and assign[perm[cursor]] >= 0:
+                    cursor += 1
+                if cursor >= n:
+                    break
+                frontier.append(int(perm[cursor]))
+            u = frontier.pop()
+            if assign[u] >= 0:
+                continue
+            assign[u] = p
+            count += 1
+            for e in range(int(indptr[u]), int(indptr[u + 1])):
+                v = int(indices[e])
+                if assign[v] < 0:
+                    frontier.append(v)
+    assign[assign < 0] = num_parts - 1
+    return assign
+
+
+# --------------------------------------------------------------------------
+# LocalGraph construction
+# --------------------------------------------------------------------------
+
+def global_degrees(g: GlobalGraph) -> Tuple[Tensor, Tensor]:
+    in_deg = torch.bincount(g.dst, minlength=g.num_nodes).float()
+    out_deg = torch.bincount(g.src, minlength=g.num_nodes).float()
+    return in_deg, out_deg
+
+
+def build_local_graph(g: GlobalGraph, assign: Tensor, rank: int,
+                      world_size: int,
+                      in_deg: Optional[Tensor] = None,
+                      out_deg: Optional[Tensor] = None) -> LocalGraph:
+    if in_deg is None or out_deg is None:
+        in_deg, out_deg = global_degrees(g)
+
+    # my in-edges: dst owned by me
+    emask = assign[g.dst] == rank
+    esrc = g.src[emask]
+    edst = g.dst[emask]
+
+    my_nodes = torch.nonzero(assign == rank, as_tuple=True)[0]   # sorted global ids
+    n_inner = int(my_nodes.numel())
+
+    # marginal = inner nodes with >=1 remote in-neighbor
+    remote_edge = assign[esrc] != rank
+    marginal_global = torch.unique(edst[remote_edge])
+    is_marginal = torch.zeros(g.num_nodes, dtype=torch.bool)
+    is_marginal[marginal_global] = True
+    central_nodes = my_nodes[~is_marginal[my_nodes]]
+    marginal_nodes = my_nodes[is_marginal[my_nodes]]
+
+    # remote nodes grouped by owner rank, sorted by global id within owner
+    remote_global = torch.unique(esrc[remote_edge])              # sorted global ids
+    remote_owner = assign[remote_global]
+    owner_order = torch.argsort(remote_owner, stable=True)       # grouped, global-sorted
+    remote_global = remote_global[owner_order]
+    recv_splits = [0] * world_size
+    for p, c in zip(*torch.unique(assign[remote_global], return_counts=True)):
+        recv_splits[int(p)] = int(c)
+
+    # local ordering: [central | marginal | remote]
+    order_global = torch.cat([central_nodes, marginal_nodes, remote_global])
+    n_central = int(central_nodes.numel())
+    n_marginal = int(marginal_nodes.numel())
+    num_nodes = int(order_global.numel())
+    g2l = torch.full((g.num_nodes,), -1, dtype=torch.int64)
+    g2l[order_global] = torch.arange(num_nodes)
+
+    lsrc = g2l[esrc]
+    ldst = g2l[edst]
+    assert int(ldst.max() if ldst.numel() else 0) < n_inner
+    indptr, indices = coo_to_csr(ldst, lsrc, n_inner)
+
+    # send_idx[q]: my nodes that q holds as remote = dsts of my in-edges with
+    # src owned by q, sorted by GLOBAL id (same order q stores them).
+    send_idx: Dict[int, Tensor] = {}
+    src_owner = assign[esrc]
+    for q in range(world_size):
+        if q == rank:
+            continue
+        mine_for_q = torch.unique(edst[src_owner == q])          # sorted global ids
+        if mine_for_q.numel():
+            send_idx[q] = g2l[mine_for_q]
+
+    lg = LocalGraph(
+        rank=rank, world_size=world_size,
+        num_central=n_central, num_marginal=n_marginal, num_nodes=num_nodes,
+        indptr=indptr, indices=indices,
+        in_deg=in_deg[order_global].clone(),
+        out_deg=out_deg[order_global].clone(),
+        send_idx=send_idx, recv_splits=recv_splits,
+        local_to_global=order_global,
+        feats=g.feats[order_global[:n_inner]].clone() if g.feats is not None else None,
+        labels=g.labels[order_global[:n_inner]].clone() if g.labels is not None else None,
+        train_mask=g.train_mask[order_global[:n_inner]].clone() if g.train_mask is not None else None,
+        val_mask=g.val_mask[order_global[:n_inner]].clone() if g.val_mask is not None else None,
+        test_mask=g.test_mask[order_global[:n_inner]].clone() if g.test_mask is not None else None,
+    )
+    lg.validate()
+    return lg
+
+
+def partition_all(g: GlobalGraph, num_parts: int, method: str = 'range',
+                  seed: int = 0) -> List[LocalGraph]:
+    if method == 'range':
+        assign = range_assignment(g.num_nodes, num_parts)
+    elif method == 'bfs':
+        assign = bfs_assignment(g, num_parts, seed)
+    else:
+        raise ValueError(f'unknown partition method {method}')
+    in_deg, out_deg = global_degrees(g)
+    return [build_local_graph(g, assign, r, num_parts, in_deg, out_deg)
+            for r in range(num_parts)]
+
+
+# --------------------------------------------------------------------------
+# on-disk layout (parity with the reference's partition artifacts)
+# --------------------------------------------------------------------------
+
+_FIELDS = ('num_central', 'num_marginal', 'num_nodes', 'indptr', 'indices',
+           'in_deg', 'out_deg', 'recv_splits', 'local_to_global', 'feats',
+           'labels', 'train_mask', 'val_mask', 'test_mask')
+
+
+def save_partitions(parts: List[LocalGraph], root: str, dataset: str,
+                    meta: Optional[dict] = None) -> str:
+    P = len(parts)
+    d = os.path.join(root, dataset, f'{P}part')
+    os.makedirs(d, exist_ok=True)
+    for lg in parts:
+        state = {f: getattr(lg, f) for f in _FIELDS}
+        state['send_idx'] = lg.send_idx
+        torch.save(state, os.path.join(d, f'part{lg.rank}.pt'))
+    info = {'dataset': dataset, 'num_parts': P}
+    info.update(meta or {})
+    with open(os.path.join(d, f'{dataset}.json'), 'w') as f:
+        json.dump(info, f, indent=2)
+    return d
+
+
+def load_partition(root: str, dataset: str, num_parts: int, rank: int) -> LocalGraph:
+    d = os.path.join(root, dataset, f'{num_parts}part')
+    state = torch.load(os.path.join(d, f'part{rank}.pt'), weights_only=False)
+    send_idx = state.pop('send_idx')
+    return LocalGraph(rank=rank, world_size=num_parts, send_idx=send_idx, **state)
