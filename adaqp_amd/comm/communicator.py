@@ -1,0 +1,142 @@
+"""Distributed communicator: RCCL (xGMI) tensor plane + gloo object plane.
+
+Reference parity: ``AdaQP/communicator/comm.py`` (gloo-only, CPU-staged
+ring-scheduled isend/irecv, ``comm.py:166-222``). MI355X redesign:
+
+- one process per GPU; ``init_process_group('cpu:gloo,cuda:nccl')`` gives
+  RCCL for CUDA tensors (boundary exchange + gradient all-reduce run
+  GPU-to-GPU over xGMI, no pinned-CPU staging) and gloo for CPU tensors
+  and object collectives (RCCL cannot carry pickled objects).
+- the reference's W-1 round ring schedule exists to pair gloo TCP
+  send/recvs; over point-to-point xGMI a single grouped
+  ``all_to_all_single`` uses every link concurrently — that one call
+  replaces ``fp_msg_exchange``/``qt_msg_exchange``.
+- gradient all-reduce is ONE flat-bucket RCCL all-reduce
+  (vs per-parameter all_reduce at ``runtime_util.py:71-77``).
+
+On a CPU-only host (unit tests, BASELINE config #1) the same code runs
+entirely on gloo.
+"""
+from __future__ import annotations
+
+import datetime
+import os
+from typing import List, Optional, Sequence
+
+import torch
+import torch.distributed as dist
+from torch import Tensor
+
+
+class Communicator:
+    ctx: Optional['Communicator'] = None
+
+    def __init__(self, backend: Optional[str] = None,
+                 init_method: str = 'env://', timeout_s: int = 600):
+        use_gpu = torch.cuda.is_available()
+        if backend is None:
+            backend = 'cpu:gloo,cuda:nccl' if use_gpu else 'gloo'
+        if not dist.is_initialized():
+            dist.init_process_group(backend, init_method=init_method,
+                                    timeout=datetime.timedelta(seconds=timeout_s))
+        self.backend = backend
+        self.rank = dist.get_rank()
+        self.world_size = dist.get_world_size()
+        if use_gpu:
+            local = int(os.environ.get('LOCAL_RANK', self.rank % max(torch.cuda.device_count(), 1)))
+            self.device = torch.device(f'cuda:{local}')
+            torch.cuda.set_device(self.device)
+        else:
+            self.device = torch.device('cpu')
+        Communicator.ctx = self
+
+    # ------------------------------------------------------------------
+    # collectives
+    # ------------------------------------------------------------------
+    def all_reduce_sum(self, t: Tensor, async_op: bool = False):
+        return dist.all_reduce(t, op=dist.ReduceOp.SUM, async_op=async_op)
+
+    def all_reduce_max(self, t: Tensor):
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        return t
+
+    def all_gather_object(self, obj) -> list:
+        out = [None] * self.world_size
+        dist.all_gather_object(out, obj)
+        return out
+
+    def broadcast_object(self, obj, src: int = 0):
+        lst = [obj]
+        dist.broadcast_object_list(lst, src=src)
+        return lst[0]
+
+    def gather_object(self, obj, dst: int = 0) -> Optional[list]:
+        out = [None] * self.world_size if self.rank == dst else None
+        dist.gather_object(obj, out, dst=dst)
+        return out
+
+    def scatter_object(self, objs: Optional[list], src: int = 0):
+        out = [None]
+        dist.scatter_object_list(out, objs if self.rank == src else None, src=src)
+        return out[0]
+
+    def barrier(self):
+        dist.barrier()
+
+    # ------------------------------------------------------------------
+    # boundary exchange (the hot path)
+    # ------------------------------------------------------------------
+    def all_to_all_v(self, out: Tensor, inp: Tensor,
+                     out_splits: Sequence[int], in_splits: Sequence[int],
+                     async_op: bool = False):
+        """Single fused variable-size all-to-all. On RCCL this is grouped
+        xGMI send/recv on the current stream; on gloo, alltoallv."""
+        return dist.all_to_all_single(out, inp, list(out_splits),
+                                      list(in_splits), async_op=async_op)
+
+    def exchange_rows(self, send: Tensor, send_splits: Sequence[int],
+                      recv_splits: Sequence[int], out: Optional[Tensor] = None,
+                      async_op: bool = False):
+        """Exchange 2-D row blocks [n_i, F] -> recv [R, F].
+
+        With the remote block stored in owner order (graph/partition.py),
+        the output of this call IS the remote feature block — no scatter.
+        """
+        F = send.shape[1] if send.dim() == 2 else 1
+        R = int(sum(recv_splits))
+        if out is None:
+            out = torch.empty((R, F) if send.dim() == 2 else (R,),
+                              dtype=send.dtype, device=send.device)
+        work = self.all_to_all_v(out, send, recv_splits, send_splits,
+                                 async_op=async_op)
+        return (out, work) if async_op else (out, None)
+
+    # ------------------------------------------------------------------
+    # gradient synchronization: one flat bucket over RCCL
+    # ------------------------------------------------------------------
+    def flat_all_reduce_grads(self, parameters) -> None:
+        grads = [p.grad for p in parameters if p.grad is not None]
+        if not grads:
+            return
+        flat = torch._utils._flatten_dense_tensors(grads)
+        dist.all_reduce(flat, op=dist.ReduceOp.SUM)
+        for g, synced in zip(grads, torch._utils._unflatten_dense_tensors(flat, grads)):
+            g.copy_(synced)
+
+    def sync_model_params(self, module: torch.nn.Module) -> None:
+        """Broadcast rank-0 weights (reference zeroes+all-reduces,
+        ``runtime_util.py:55-63``; broadcast is the direct form)."""
+        for p in module.parameters():
+            dist.broadcast(p.data, src=0)
+
+    def sync_seed(self, seed: Optional[int] = None) -> int:
+        seed = self.broadcast_object(seed if seed is not None else
+                                     int(torch.randint(0, 2**31 - 1, (1,)).item()))
+        torch.manual_seed(seed)
+        return seed
+
+    @staticmethod
+    def shutdown():
+        if dist.is_initialized():
+            dist.destroy_process_group()
+        Communicator.ctx = None

@@ -1,0 +1,61 @@
+"""DistGCN: aggregate-then-transform GCN over partitioned graphs.
+
+Reference parity: ``AdaQP/model/distGCN.py`` (DistGCNConv aggregates with
+DistAggConv then applies the linear transform; stack uses
+dropout -> LayerNorm -> ReLU between layers; Xavier init)."""
+from __future__ import annotations
+
+import math
+from typing import List
+
+import torch
+import torch.nn as nn
+from torch import Tensor
+
+from ..ops.dist_agg import dist_aggregate
+
+
+class DistGCNConv(nn.Module):
+    def __init__(self, in_dim: int, out_dim: int, layer: int, use_bias: bool = True):
+        super().__init__()
+        self.layer = layer
+        self.weight = nn.Parameter(torch.empty(in_dim, out_dim))
+        self.bias = nn.Parameter(torch.zeros(out_dim)) if use_bias else None
+        self.reset_parameters()
+
+    def reset_parameters(self):
+        nn.init.xavier_uniform_(self.weight)
+        if self.bias is not None:
+            nn.init.zeros_(self.bias)
+
+    def forward(self, engine, x: Tensor) -> Tensor:
+        rst = dist_aggregate(x, engine, self.layer, self.training)
+        rst = rst @ self.weight
+        if self.bias is not None:
+            rst = rst + self.bias
+        return rst
+
+
+class DistGCN(nn.Module):
+    def __init__(self, in_dim: int, hidden_dim: int, out_dim: int,
+                 num_layers: int = 3, dropout: float = 0.5,
+                 use_norm: bool = True):
+        super().__init__()
+        dims = [in_dim] + [hidden_dim] * (num_layers - 1) + [out_dim]
+        self.convs = nn.ModuleList(
+            [DistGCNConv(dims[i], dims[i + 1], layer=i) for i in range(num_layers)])
+        self.norms = nn.ModuleList(
+            [nn.LayerNorm(hidden_dim, elementwise_affine=True)
+             for _ in range(num_layers - 1)]) if use_norm else None
+        self.dropout = nn.Dropout(dropout)
+
+    def forward(self, engine, feats: Tensor) -> Tensor:
+        h = feats
+        for i, conv in enumerate(self.convs):
+            h = self.dropout(h) if i > 0 else h
+            h = conv(engine, h)
+            if i < len(self.convs) - 1:
+                if self.norms is not None:
+                    h = self.norms[i](h)
+                h = torch.relu(h)
+        return h

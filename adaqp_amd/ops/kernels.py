@@ -1,0 +1,138 @@
+"""Device dispatch for the hot ops.
+
+GPU (MI355X): hand-written CDNA4 HIP kernels from ``adaqp_amd/csrc``
+(built in-tree as ``adaqp_amd._C``). The GPU path REFUSES to fall back —
+if the extension is missing on a CUDA device, that is a build error, not
+a reason to silently run eager PyTorch.
+
+CPU: torch reference implementations (bit-exact for quant pack/unpack —
+same hash RNG) used by unit tests and the CPU/gloo plumbing config.
+"""
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+from torch import Tensor
+
+from . import quant as Q
+from ..comm.buffers import SidePlan, BITS_SET, bytes_per_node
+
+_native = None
+_native_err: Optional[Exception] = None
+
+
+def native():
+    global _native, _native_err
+    if _native is None and _native_err is None:
+        try:
+            from .. import _C  # in-tree HIP extension
+            _native = _C
+        except Exception as e:   # pragma: no cover
+            _native_err = e
+    if _native is None:
+        raise RuntimeError(
+            f'adaqp_amd._C HIP extension not built (run `python setup.py '
+            f'build_ext --inplace`): {_native_err}')
+    return _native
+
+
+def has_native() -> bool:
+    try:
+        native()
+        return True
+    except RuntimeError:
+        return False
+
+
+# --------------------------------------------------------------------------
+# mixed-bit quantize / dequantize
+# --------------------------------------------------------------------------
+
+def mixed_quantize(x: Tensor, plan: SidePlan, seed: int) -> Tuple[Tensor, Tensor]:
+    """Gather plan.rows from x, quantize per bit group, pack into the wire
+    payload + params described by plan. Returns (uint8 [B], bf16 [2S])."""
+    if x.is_cuda:
+        payload = torch.empty(plan.total_bytes, dtype=torch.uint8, device=x.device)
+        params = torch.empty(2 * plan.total_nodes, dtype=torch.bfloat16, device=x.device)
+        for b in BITS_SET:
+            rows = plan.rows[b]
+            if rows.numel():
+                native().quant_pack(x, rows, plan.pos[b], plan.off[b], b,
+                                    seed, payload, params)
+        return payload, params
+    payload = torch.zeros(plan.total_bytes, dtype=torch.uint8)
+    params = torch.zeros(2 * plan.total_nodes, dtype=torch.bfloat16)
+    for b in BITS_SET:
+        rows, pos, off = plan.rows[b], plan.pos[b], plan.off[b]
+        if rows.numel() == 0:
+            continue
+        pl, scale, rmin = Q.pack_torch(x[rows].float(), b, seed, node_tag=pos)
+        bpn = pl.shape[1]
+        idx = off[:, None] + torch.arange(bpn, dtype=torch.int64)
+        payload[idx.reshape(-1)] = pl.reshape(-1)
+        params[2 * pos] = scale
+        params[2 * pos + 1] = rmin
+    return payload, params
+
+
+def mixed_dequantize(payload: Tensor, params: Tensor, plan: SidePlan,
+                     out: Tensor) -> Tensor:
+    """Unpack wire payload into fp rows of ``out`` (remote block [R, F]),
+    fused with the scatter to plan.rows."""
+    if payload.is_cuda:
+        for b in BITS_SET:
+            rows = plan.rows[b]
+            if rows.numel():
+                native().quant_unpack(payload, params, rows, plan.pos[b],
+                                      plan.off[b], b, plan.F, out)
+        return out
+    for b in BITS_SET:
+        rows, pos, off = plan.rows[b], plan.pos[b], plan.off[b]
+        if rows.numel() == 0:
+            continue
+        bpn = bytes_per_node(plan.F, b)
+        idx = off[:, None] + torch.arange(bpn, dtype=torch.int64)
+        pl = payload[idx.reshape(-1)].reshape(-1, bpn)
+        scale = params[2 * pos].clone()
+        rmin = params[2 * pos + 1].clone()
+        out[rows] = Q.unpack_torch(pl, b, scale, rmin, plan.F).to(out.dtype)
+    return out
+
+
+# --------------------------------------------------------------------------
+# CSR SpMM:  y[r] = dst_scale[r] * sum_{c in row r} src_scale[c] * x[c]
+# --------------------------------------------------------------------------
+
+_csr_cache: dict = {}
+
+
+def spmm(indptr: Tensor, indices: Tensor, x: Tensor,
+         src_scale: Optional[Tensor], dst_scale: Optional[Tensor],
+         num_rows: int) -> Tensor:
+    """Aggregation SpMM over an in-edge CSR.
+
+    x: [N, F] full feature matrix (local rows first, then remote).
+    src_scale: [N] or None; dst_scale: [num_rows] or None.
+    """
+    if x.is_cuda:
+        out = torch.empty(num_rows, x.shape[1], dtype=x.dtype, device=x.device)
+        native().spmm_csr(indptr, indices, x, out,
+                          src_scale if src_scale is not None else torch.empty(0, device=x.device),
+                          dst_scale if dst_scale is not None else torch.empty(0, device=x.device))
+        return out
+    xs = x * src_scale[:, None] if src_scale is not None else x
+    key = (indptr.data_ptr(), indices.data_ptr(), num_rows, x.shape[0])
+    sp = _csr_cache.get(key)
+    if sp is None:
+        crow = indptr.to(torch.int64)
+        col = indices.to(torch.int64)
+        val = torch.ones(col.numel(), dtype=torch.float32)
+        sp = torch.sparse_csr_tensor(crow, col, val, size=(num_rows, x.shape[0]))
+        if len(_csr_cache) > 64:
+            _csr_cache.clear()
+        _csr_cache[key] = sp
+    y = torch.sparse.mm(sp, xs.float()).to(x.dtype)
+    if dst_scale is not None:
+        y = y * dst_scale[:, None]
+    return y

@@ -1,0 +1,136 @@
+"""GraphEngine: per-rank runtime state (graph, norms, streams, plans).
+
+Reference parity: ``AdaQP/manager/graphEngine.py`` (GraphEngine +
+DecompGraph + stream/thread/event context). MI355X redesign:
+
+- the central/marginal decomposition is a row-range split of one CSR
+  (graph/csr.py), so ``DecompGraph`` and its per-layer copy buffers do
+  not exist.
+- overlap state is ONE side HIP stream + cuda events; no CPU thread, no
+  CPU events (the reference needs them because gloo runs on the host;
+  RCCL collectives are stream-ordered device work,
+  ``graphEngine.py:122-132`` -> ``ops/dist_agg.py`` here).
+- normalization vectors for every (model, direction) pair are
+  precomputed once; backward reuses the forward CSR with swapped scales
+  (valid for the bidirected graphs this framework targets — every
+  reference dataset; asserted at load).
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Optional
+
+import torch
+from torch import Tensor
+
+from ..graph import LocalGraph
+from ..helpers import BitType, DistGNNType, RunMode
+from ..comm.buffers import KeyPlan, build_key_plan, uniform_bits
+from ..comm.communicator import Communicator
+from .timer import Timer
+
+
+class GraphEngine:
+    ctx: Optional['GraphEngine'] = None
+
+    def __init__(self, graph: LocalGraph, mode: RunMode,
+                 model_type: DistGNNType, msg_dims: List[int],
+                 agg_type: str = 'mean', device: Optional[torch.device] = None,
+                 base_seed: int = 2026):
+        self.device = device if device is not None else (
+            Communicator.ctx.device if Communicator.ctx else torch.device('cpu'))
+        self.graph = graph.to(self.device)
+        self.mode = mode
+        self.bit_type = mode.bit_type
+        self.use_parallel = mode.use_parallel
+        self.model_type = model_type
+        self.agg_type = agg_type
+        self.msg_dims = list(msg_dims)          # feature dim per layer exchange
+        self.num_layers = len(msg_dims)
+        self.base_seed = base_seed
+        self._rng_counter = 0
+        self.timer = Timer(enabled=False, cuda=self.device.type == 'cuda')
+        self.is_tracing = False
+        self.traced: Dict[str, Tensor] = {}      # key -> accumulated per-send-node variance proxy
+
+        g = self.graph
+        din = g.in_deg.clamp(min=1.0)
+        dout = g.out_deg.clamp(min=1.0)
+        I = g.num_inner
+        # GCN: y = Din^-1/2 A Dout^-1/2 x ; backward identical with swapped
+        # roles (symmetric graph -> same vectors)
+        self.gcn_src_f = dout.pow(-0.5)          # [N]
+        self.gcn_dst_f = din.pow(-0.5)[:I]       # [I]
+        self.gcn_src_b = din.pow(-0.5)           # [N]
+        self.gcn_dst_b = dout.pow(-0.5)[:I]
+        # SAGE mean: y = Din^-1 A x ; backward: g_x = A^T Din^-1 g
+        self.sage_dst_f = din.pow(-1.0)[:I]
+        self.sage_src_b = din.pow(-1.0)
+        # SAGE gcn: y = (A x + x) / (Din + 1)
+        self.sage1_dst_f = (din + 1.0).pow(-1.0)[:I]
+        self.sage1_src_b = (din + 1.0).pow(-1.0)
+
+        # decomposition views (zero-copy)
+        self.full_view = (g.indptr, g.indices, 0, g.num_inner)
+        cptr, cidx, _ = g.central_view()
+        self.central_view = (cptr, cidx, 0, g.num_central)
+        mptr, midx, mbase = g.marginal_view()
+        self.marginal_view = (mptr, midx, mbase, g.num_marginal)
+
+        # overlap streams/events
+        if self.device.type == 'cuda':
+            self.comm_stream = torch.cuda.Stream(device=self.device)
+            self.remote_ready = torch.cuda.Event()
+        else:
+            self.comm_stream = None
+            self.remote_ready = None
+
+        # quantization plans per exchange key
+        self.plans: Dict[str, KeyPlan] = {}
+        self.assignments: Dict[str, Dict[int, Tensor]] = {}
+        GraphEngine.ctx = self
+
+    # ------------------------------------------------------------------
+    def exchange_keys(self) -> List[str]:
+        keys = [f'forward{l}' for l in range(self.num_layers)]
+        keys += [f'backward{l}' for l in range(1, self.num_layers)]
+        return keys
+
+    def key_dim(self, key: str) -> int:
+        return self.msg_dims[int(key.replace('forward', '').replace('backward', ''))]
+
+    def set_assignment(self, assignments: Dict[str, Dict[int, Tensor]]) -> None:
+        """assignments: key -> {peer -> int64 bits over send_idx[peer] order}.
+        Rebuilds all wire plans (reference: CommBuffer._update,
+        ``buffer.py:255-264``)."""
+        comm = Communicator.ctx
+        self.assignments = assignments
+        self.plans = {}
+        for key in self.exchange_keys():
+            self.plans[key] = build_key_plan(
+                comm, self.graph, self.key_dim(key), assignments[key],
+                device=self.device)
+
+    def set_uniform_assignment(self, bits: int = 8) -> None:
+        self.set_assignment({k: uniform_bits(self.graph, bits)
+                             for k in self.exchange_keys()})
+
+    def next_seed(self) -> int:
+        self._rng_counter += 1
+        return (self.base_seed + 1000003 * self._rng_counter) & 0x7FFFFFFF
+
+    # ---- variance tracing for the adaptive assigner ----
+    def trace(self, key: str, send: Tensor) -> None:
+        """Accumulate the per-node quantization-variance proxy
+        (dim/6)*(rmax-rmin)^2 (reference ``op_util.py:91-99``)."""
+        if send.numel() == 0:
+            return
+        with torch.no_grad():
+            rng = send.max(dim=1).values - send.min(dim=1).values
+            v = (send.shape[1] / 6.0) * rng.float() ** 2
+            if key in self.traced:
+                self.traced[key] += v
+            else:
+                self.traced[key] = v.clone()
+
+    def reset_trace(self) -> None:
+        self.traced = {}
