@@ -1,0 +1,293 @@
+// CDNA4 (gfx950 / MI355X) HIP kernels for adaqp_amd.
+//
+// Reference parity: quant_cuda pack/unpack
+// (/root/reference/AdaQP/util/quantization/src/quantization_cuda_kernel.cu:34-156)
+// and DGL's update_all CSR SpMM (delegated by the reference at
+// AdaQP/model/ops.py:30). MI355X-first redesign:
+//
+//  * quant_pack fuses per-node min/max + scale + stochastic round + bit-pack
+//    + the per-(peer,bit-group) wire scatter into ONE pass (the reference
+//    computes rmin/rmax with two torch reductions in Python first).
+//  * packing is along the FEATURE axis so each node is a contiguous byte
+//    run: lane l owns (8/bits) consecutive features and emits whole bytes —
+//    coalesced float4/float2/float loads and byte stores across the
+//    64-wide wavefront (the reference packs along the node axis).
+//  * RNG is a stateless counter hash (triple32) of (seed, node, feature) —
+//    no curand state; bit-compatible with the CPU oracle in ops/quant.py.
+//  * spmm_csr: one 64-lane wavefront per destination row, lane l owning
+//    features [4l, 4l+4) (float4), with fused src/dst degree scaling.
+//
+// Built for gfx950 only. No CUDA compatibility path.
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include <cstdint>
+
+#define WAVE 64
+#define CHECK_DEV(x) TORCH_CHECK(x.is_cuda(), #x " must be on the GPU")
+#define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
+
+// ---------------------------------------------------------------------------
+// RNG: triple32 hash -> U[0,1). Must match ops/quant.py::_hash_u32 exactly.
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ uint32_t hash_u32(uint32_t h) {
+    h ^= h >> 16; h *= 0x7FEB352Du;
+    h ^= h >> 15; h *= 0x846CA68Bu;
+    h ^= h >> 16;
+    return h;
+}
+
+__device__ __forceinline__ float uniform01(uint32_t seed, uint32_t tag, uint32_t feat) {
+    uint32_t h = seed ^ (tag * 0x9E3779B9u) ^ (feat * 0x85EBCA6Bu);
+    return (float)((double)hash_u32(h) * 2.3283064365386963e-10);
+}
+
+// round-to-nearest-even f32 -> bf16 bits (parity with torch .to(bfloat16))
+__device__ __forceinline__ uint16_t f32_to_bf16(float f) {
+    uint32_t b = __float_as_uint(f);
+    uint32_t r = (b + 0x7FFFu + ((b >> 16) & 1u)) >> 16;
+    return (uint16_t)r;
+}
+__device__ __forceinline__ float bf16_to_f32(uint16_t h) {
+    return __uint_as_float(((uint32_t)h) << 16);
+}
+
+// ---------------------------------------------------------------------------
+// quant_pack: one wave per node.
+//   rows[i]   : row of x to quantize
+//   pos[i]    : wire node position (params slot + RNG tag)
+//   off[i]    : wire byte offset of this node's payload
+// Lane l owns features [l*vpb + k*WAVE*vpb, ...): it loads vpb consecutive
+// floats, reduces min/max across the wave, then emits one byte per k.
+// ---------------------------------------------------------------------------
+template <int BITS>
+__global__ void quant_pack_kernel(
+    const float* __restrict__ x, const int64_t* __restrict__ rows,
+    const int64_t* __restrict__ pos, const int64_t* __restrict__ off,
+    int64_t n, int64_t F, int64_t ld, uint32_t seed,
+    uint8_t* __restrict__ payload, uint16_t* __restrict__ params) {
+    constexpr int VPB = 8 / BITS;          // values per byte
+    const int64_t wid = (int64_t)blockIdx.x * (blockDim.x / WAVE)
+                      + threadIdx.x / WAVE;
+    if (wid >= n) return;
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int64_t row = rows[wid];
+    const float* xr = x + row * ld;
+    const int64_t bpn = (F * BITS + 7) / 8;
+
+    // pass 1: min/max (coalesced vpb-wide vector loads)
+    float mn = 1e38f, mx = -1e38f;
+    for (int64_t f0 = (int64_t)lane * VPB; f0 < F; f0 += (int64_t)WAVE * VPB) {
+#pragma unroll
+        for (int k = 0; k < VPB; ++k) {
+            if (f0 + k < F) {
+                float v = xr[f0 + k];
+                mn = fminf(mn, v);
+                mx = fmaxf(mx, v);
+            }
+        }
+    }
+#pragma unroll
+    for (int o = 32; o; o >>= 1) {
+        mn = fminf(mn, __shfl_xor(mn, o));
+        mx = fmaxf(mx, __shfl_xor(mx, o));
+    }
+    const float rng = mx - mn;
+    float scale_f = rng > 0.f ? ((float)((1 << BITS) - 1)) / fmaxf(rng, 1e-30f) : 0.f;
+    const uint16_t scale_h = f32_to_bf16(scale_f);
+    const uint16_t rmin_h = f32_to_bf16(mn);
+    const float scale = bf16_to_f32(scale_h);     // quantize with the wire scale
+    const float rmin = bf16_to_f32(rmin_h);
+    const int64_t p = pos[wid];
+    if (lane == 0) {
+        params[2 * p] = scale_h;
+        params[2 * p + 1] = rmin_h;
+    }
+    const uint32_t tag = (uint32_t)p;
+
+    // pass 2: quantize + pack one byte per lane-step
+    uint8_t* out = payload + off[wid];
+    for (int64_t b0 = lane; b0 * VPB < F; b0 += WAVE) {
+        uint32_t byte = 0;
+#pragma unroll
+        for (int k = 0; k < VPB; ++k) {
+            const int64_t f = b0 * VPB + k;
+            if (f < F && scale > 0.f) {
+                float v = (xr[f] - rmin) * scale;
+                float u = uniform01(seed, tag, (uint32_t)f);
+                int q = (int)floorf(v + u);
+                q = max(0, min(q, (1 << BITS) - 1));
+                byte |= ((uint32_t)q) << (k * BITS);
+            }
+        }
+        out[b0] = (uint8_t)byte;
+    }
+}
+
+// ---------------------------------------------------------------------------
+// quant_unpack: one wave per node, fused scatter into out[rows[i]].
+// ---------------------------------------------------------------------------
+template <int BITS>
+__global__ void quant_unpack_kernel(
+    const uint8_t* __restrict__ payload, const uint16_t* __restrict__ params,
+    const int64_t* __restrict__ rows, const int64_t* __restrict__ pos,
+    const int64_t* __restrict__ off, int64_t n, int64_t F, int64_t ld,
+    float* __restrict__ out) {
+    constexpr int VPB = 8 / BITS;
+    const int64_t wid = (int64_t)blockIdx.x * (blockDim.x / WAVE)
+                      + threadIdx.x / WAVE;
+    if (wid >= n) return;
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int64_t p = pos[wid];
+    const float scale = bf16_to_f32(params[2 * p]);
+    const float rmin = bf16_to_f32(params[2 * p + 1]);
+    const float inv = scale > 0.f ? 1.f / scale : 0.f;
+    const uint8_t* in = payload + off[wid];
+    float* o = out + rows[wid] * ld;
+    for (int64_t b0 = lane; b0 * VPB < F; b0 += WAVE) {
+        const uint32_t byte = in[b0];
+#pragma unroll
+        for (int k = 0; k < VPB; ++k) {
+            const int64_t f = b0 * VPB + k;
+            if (f < F) {
+                const int q = (byte >> (k * BITS)) & ((1 << BITS) - 1);
+                o[f] = (float)q * inv + rmin;
+            }
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// spmm_csr: y[r] = dst_scale[r] * sum_{e in row r} src_scale[c_e] * x[c_e]
+// One wave per row, grid-stride; lane l owns features 4l..4l+3 (float4)
+// and loops over feature chunks of 256.
+// ---------------------------------------------------------------------------
+__global__ void spmm_csr_kernel(
+    const int64_t* __restrict__ indptr, const int64_t* __restrict__ indices,
+    const float* __restrict__ x, float* __restrict__ y,
+    const float* __restrict__ src_scale, const float* __restrict__ dst_scale,
+    int64_t num_rows, int64_t F) {
+    const int waves_per_block = blockDim.x / WAVE;
+    const int64_t wid0 = (int64_t)blockIdx.x * waves_per_block + threadIdx.x / WAVE;
+    const int64_t stride = (int64_t)gridDim.x * waves_per_block;
+    const int lane = threadIdx.x & (WAVE - 1);
+
+    for (int64_t r = wid0; r < num_rows; r += stride) {
+        const int64_t e0 = indptr[r], e1 = indptr[r + 1];
+        const float ds = dst_scale ? dst_scale[r] : 1.f;
+        for (int64_t f0 = (int64_t)lane * 4; f0 < F; f0 += WAVE * 4) {
+            float4 acc = {0.f, 0.f, 0.f, 0.f};
+            const bool full = (f0 + 4 <= F);
+            for (int64_t e = e0; e < e1; ++e) {
+                const int64_t c = indices[e];
+                const float s = src_scale ? src_scale[c] : 1.f;
+                const float* xc = x + c * F + f0;
+                if (full) {
+                    const float4 v = *reinterpret_cast<const float4*>(xc);
+                    acc.x = fmaf(v.x, s, acc.x);
+                    acc.y = fmaf(v.y, s, acc.y);
+                    acc.z = fmaf(v.z, s, acc.z);
+                    acc.w = fmaf(v.w, s, acc.w);
+                } else {
+                    for (int k = 0; k < 4 && f0 + k < F; ++k)
+                        ((float*)&acc)[k] = fmaf(xc[k], s, ((float*)&acc)[k]);
+                }
+            }
+            float* yr = y + r * F + f0;
+            if (full) {
+                acc.x *= ds; acc.y *= ds; acc.z *= ds; acc.w *= ds;
+                *reinterpret_cast<float4*>(yr) = acc;
+            } else {
+                for (int k = 0; k < 4 && f0 + k < F; ++k)
+                    yr[k] = ((float*)&acc)[k] * ds;
+            }
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// host launchers
+// ---------------------------------------------------------------------------
+static inline hipStream_t cur_stream() {
+    return at::cuda::getCurrentHIPStream().stream();
+}
+
+void quant_pack(torch::Tensor x, torch::Tensor rows, torch::Tensor pos,
+                torch::Tensor off, int64_t bits, int64_t seed,
+                torch::Tensor payload, torch::Tensor params) {
+    CHECK_DEV(x); CHECK_CONTIG(x); CHECK_DEV(rows); CHECK_DEV(payload); CHECK_DEV(params);
+    TORCH_CHECK(x.scalar_type() == torch::kFloat32, "quant_pack expects fp32");
+    const int64_t n = rows.numel();
+    if (n == 0) return;
+    const int64_t F = x.size(1), ld = x.stride(0);
+    const int waves_per_block = 4;
+    const dim3 block(WAVE * waves_per_block);
+    const dim3 grid((n + waves_per_block - 1) / waves_per_block);
+    auto s = cur_stream();
+#define LAUNCH(B) quant_pack_kernel<B><<<grid, block, 0, s>>>( \
+        x.data_ptr<float>(), rows.data_ptr<int64_t>(), pos.data_ptr<int64_t>(), \
+        off.data_ptr<int64_t>(), n, F, ld, (uint32_t)seed, \
+        payload.data_ptr<uint8_t>(), reinterpret_cast<uint16_t*>(params.data_ptr<at::BFloat16>()))
+    switch (bits) {
+        case 2: LAUNCH(2); break;
+        case 4: LAUNCH(4); break;
+        case 8: LAUNCH(8); break;
+        default: TORCH_CHECK(false, "bits must be 2/4/8");
+    }
+#undef LAUNCH
+}
+
+void quant_unpack(torch::Tensor payload, torch::Tensor params, torch::Tensor rows,
+                  torch::Tensor pos, torch::Tensor off, int64_t bits, int64_t F,
+                  torch::Tensor out) {
+    CHECK_DEV(payload); CHECK_DEV(out); CHECK_CONTIG(out);
+    TORCH_CHECK(out.scalar_type() == torch::kFloat32, "quant_unpack expects fp32 out");
+    const int64_t n = rows.numel();
+    if (n == 0) return;
+    const int waves_per_block = 4;
+    const dim3 block(WAVE * waves_per_block);
+    const dim3 grid((n + waves_per_block - 1) / waves_per_block);
+    auto s = cur_stream();
+#define LAUNCH(B) quant_unpack_kernel<B><<<grid, block, 0, s>>>( \
+        payload.data_ptr<uint8_t>(), \
+        reinterpret_cast<uint16_t*>(params.data_ptr<at::BFloat16>()), \
+        rows.data_ptr<int64_t>(), pos.data_ptr<int64_t>(), off.data_ptr<int64_t>(), \
+        n, F, out.stride(0), out.data_ptr<float>())
+    switch (bits) {
+        case 2: LAUNCH(2); break;
+        case 4: LAUNCH(4); break;
+        case 8: LAUNCH(8); break;
+        default: TORCH_CHECK(false, "bits must be 2/4/8");
+    }
+#undef LAUNCH
+}
+
+void spmm_csr(torch::Tensor indptr, torch::Tensor indices, torch::Tensor x,
+              torch::Tensor y, torch::Tensor src_scale, torch::Tensor dst_scale) {
+    CHECK_DEV(x); CHECK_CONTIG(x); CHECK_DEV(y); CHECK_CONTIG(y);
+    TORCH_CHECK(x.scalar_type() == torch::kFloat32, "spmm_csr expects fp32");
+    const int64_t num_rows = y.size(0), F = x.size(1);
+    TORCH_CHECK(indptr.numel() == num_rows + 1, "indptr/num_rows mismatch");
+    const int waves_per_block = 4;
+    const dim3 block(WAVE * waves_per_block);
+    // >> 256 workgroups to fill 8 XCDs; cap and grid-stride
+    int64_t blocks = (num_rows + waves_per_block - 1) / waves_per_block;
+    blocks = std::min<int64_t>(blocks, 8192);
+    const dim3 grid(std::max<int64_t>(blocks, 1));
+    spmm_csr_kernel<<<grid, block, 0, cur_stream()>>>(
+        indptr.data_ptr<int64_t>(), indices.data_ptr<int64_t>(),
+        x.data_ptr<float>(), y.data_ptr<float>(),
+        src_scale.numel() ? src_scale.data_ptr<float>() : nullptr,
+        dst_scale.numel() ? dst_scale.data_ptr<float>() : nullptr,
+        num_rows, F);
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("quant_pack", &quant_pack,
+          "fused minmax+stochastic-quantize+bitpack (CDNA4)");
+    m.def("quant_unpack", &quant_unpack,
+          "fused dequantize+scatter (CDNA4)");
+    m.def("spmm_csr", &spmm_csr,
+          "CSR SpMM with fused degree normalization (CDNA4)");
+}

@@ -1,0 +1,164 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: ogbn-products-shaped GraphSAGE, AdaQP mode.
+
+Driver contract: ``python bench.py --gpus N --steps K --warmup W`` runs
+one rank per GPU (launched via torch.distributed.run for N>1), does W
+untimed warmup epochs, times EXACTLY K epochs bracketed by
+barrier + torch.cuda.synchronize on both sides, takes the MAX over
+ranks, and rank 0 prints ONE JSON line.
+
+Metric (BASELINE.json): per-epoch time (s), ogbn-products GraphSAGE
+8-part, on a synthetic graph of that shape (no network for datasets)
+with random-init weights. Strong scaling: the global graph is fixed;
+N ranks = N partitions.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument('--gpus', type=int, default=1)
+    p.add_argument('--steps', type=int, default=10)
+    p.add_argument('--warmup', type=int, default=3)
+    p.add_argument('--mode', type=str, default='AdaQP',
+                   choices=['Vanilla', 'AdaQP', 'AdaQP-q', 'AdaQP-p'])
+    p.add_argument('--model', type=str, default='sage', choices=['gcn', 'sage'])
+    p.add_argument('--dataset', type=str, default='ogbn-products')
+    p.add_argument('--scale', type=float, default=1.0,
+                   help='shrink the synthetic graph (debug only)')
+    p.add_argument('--hidden', type=int, default=256)
+    p.add_argument('--layers', type=int, default=3)
+    p.add_argument('--assign-bits', type=int, default=4,
+                   help='uniform bit width for quantized modes')
+    p.add_argument('--part-dir', type=str, default='part_data_bench')
+    p.add_argument('--cpu', action='store_true', help='force CPU (debug)')
+    return p.parse_args()
+
+
+def get_dist_env(args):
+    rank = int(os.environ.get('RANK', 0))
+    world = int(os.environ.get('WORLD_SIZE', 1))
+    if world == 1 and args.gpus > 1:
+        raise SystemExit('launch N>1 via torch.distributed.run')
+    os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
+    os.environ.setdefault('MASTER_PORT', '29671')
+    os.environ.setdefault('RANK', '0')
+    os.environ.setdefault('WORLD_SIZE', '1')
+    os.environ.setdefault('LOCAL_RANK', str(rank))
+    return rank, world
+
+
+def prepare_partition(args, rank, world):
+    """Rank 0 generates+partitions the synthetic graph once; all load."""
+    from adaqp_amd.graph import synth_graph, partition_all, save_partitions, load_partition
+    import torch.distributed as dist
+    tag = f'{args.dataset}_s{args.scale}'
+    d = os.path.join(args.part_dir, tag, f'{world}part')
+    marker = os.path.join(d, f'{tag}.json')
+    if rank == 0 and not os.path.exists(marker):
+        t0 = time.time()
+        g = synth_graph(args.dataset, world, seed=17, scale=args.scale)
+        parts = partition_all(g, world, method='range')
+        save_partitions(parts, args.part_dir, tag)
+        print(f'# partition prep {time.time()-t0:.1f}s: '
+              f'{g.num_nodes} nodes {g.num_edges} edges', file=sys.stderr)
+    if world > 1:
+        dist.barrier()
+    return load_partition(args.part_dir, tag, world, rank)
+
+
+def main():
+    args = parse_args()
+    rank, world = get_dist_env(args)
+
+    from adaqp_amd.comm import Communicator
+    from adaqp_amd.runtime import GraphEngine
+    from adaqp_amd.runtime.utils import train_epoch, global_train_count, evaluate
+    from adaqp_amd.models import DistGCN, DistSAGE
+    from adaqp_amd.helpers import RunMode, DistGNNType
+    from adaqp_amd.graph import DATASET_SHAPES
+
+    use_gpu = torch.cuda.is_available() and not args.cpu
+    comm = Communicator(backend=None if use_gpu else 'gloo')
+    lg = prepare_partition(args, rank, world)
+
+    _, _, feat_dim, num_classes, multilabel = DATASET_SHAPES[args.dataset]
+    mode = RunMode(args.mode)
+    mtype = DistGNNType.DistGCN if args.model == 'gcn' else DistGNNType.DistSAGE
+    msg_dims = [feat_dim] + [args.hidden] * (args.layers - 1)
+    engine = GraphEngine(lg, mode, mtype, msg_dims, agg_type='mean',
+                         device=comm.device)
+    if mode.bit_type.name == 'QUANT':
+        engine.set_uniform_assignment(args.assign_bits)
+
+    torch.manual_seed(12345)
+    if args.model == 'gcn':
+        model = DistGCN(feat_dim, args.hidden, num_classes, args.layers)
+    else:
+        model = DistSAGE(feat_dim, args.hidden, num_classes, args.layers,
+                         aggregator_type='mean')
+    model = model.to(comm.device)
+    comm.sync_model_params(model)
+    opt = torch.optim.Adam(model.parameters(), lr=0.01)
+    gc = global_train_count(engine)
+
+    def sync():
+        comm.barrier()
+        if use_gpu:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        train_epoch(engine, model, opt, gc, multilabel)
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        train_epoch(engine, model, opt, gc, multilabel)
+    sync()
+    elapsed = torch.tensor([time.perf_counter() - t0],
+                           device=comm.device if use_gpu else 'cpu')
+    comm.all_reduce_max(elapsed)
+    per_epoch = float(elapsed.item()) / args.steps
+
+    metrics = evaluate(engine, model, multilabel)
+
+    if rank == 0:
+        n, e, _, _, _ = DATASET_SHAPES[args.dataset]
+        out = {
+            'metric': 'per-epoch time (s), ogbn-products GraphSAGE 8-part'
+                      if args.dataset == 'ogbn-products' and args.model == 'sage'
+                      else f'per-epoch time (s), {args.dataset} {args.model} {world}-part',
+            'value': per_epoch,
+            'unit': 's/epoch',
+            'n_gpus': world,
+            'steps': args.steps,
+            'warmup': args.warmup,
+            'ms_per_step': per_epoch * 1000.0,
+            'higher_is_better': False,
+            'scaling': 'strong',
+            'vs_baseline': None,
+            'dtype': 'fp32',
+            'data': f'synthetic ({args.dataset} shape: {int(n*args.scale)} nodes, '
+                    f'~{int(e*args.scale)} edges, planted METIS-like locality), random-init weights',
+            'config': {
+                'model': f'{"GraphSAGE" if args.model == "sage" else "GCN"}'
+                         f' 3x{args.hidden}',
+                'dataset': args.dataset,
+                'mode': args.mode,
+                'global_batch': 'full-graph',
+                'seq_len': None,
+                'parallelism': f'graph-partition dp{world}',
+                'test_acc': metrics['test'],
+            },
+        }
+        print(json.dumps(out))
+    Communicator.shutdown()
+
+
+if __name__ == '__main__':
+    main()
