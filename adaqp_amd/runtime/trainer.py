@@ -1,0 +1,174 @@
+"""Trainer: orchestration of one training run.
+
+Reference parity: ``AdaQP/trainer/trainer.py`` (config load + runtime
+override, subsystem construction order, epoch loop with periodic bit
+re-assignment, metric/time CSV artifacts under ``exp/``).
+
+Construction order mirrors the reference (``trainer.py:50-71``):
+Communicator -> GraphEngine -> Assigner -> initial assignment -> Model.
+"""
+from __future__ import annotations
+
+import logging
+import os
+import time
+from typing import Dict, Optional
+
+import torch
+import yaml
+
+from ..comm.communicator import Communicator
+from ..graph import (DATASET_SHAPES, load_partition, partition_all,
+                     save_partitions, synth_graph)
+from ..helpers import AssignScheme, DistGNNType, RunMode
+from ..models import DistGCN, DistSAGE
+from ..assigner import Assigner
+from .engine import GraphEngine
+from .recorder import Recorder
+from .utils import evaluate, global_train_count, train_epoch
+
+CONFIG_DIR = os.path.join(os.path.dirname(__file__), '..', 'config')
+MODEL_MAP = {'gcn': DistGNNType.DistGCN, 'sage': DistGNNType.DistSAGE}
+
+logger = logging.getLogger('trainer')
+
+
+def load_config(dataset: str) -> dict:
+    with open(os.path.join(CONFIG_DIR, 'synthetic.yaml')) as f:
+        cfg = yaml.safe_load(f)
+    p = os.path.join(CONFIG_DIR, f'{dataset}.yaml')
+    if os.path.exists(p):
+        with open(p) as f:
+            override = yaml.safe_load(f) or {}
+        for sec, vals in override.items():
+            cfg.setdefault(sec, {}).update(vals or {})
+    return cfg
+
+
+class Trainer:
+    def __init__(self, args):
+        self.args = args
+        self.cfg = load_config(args.dataset)
+        rt = self.cfg['runtime']
+        for k in ('num_epochs', 'lr', 'log_steps'):
+            v = getattr(args, k, None)
+            if v is not None:
+                rt[k] = v
+        self.mode = RunMode(args.mode)
+        self.model_type = MODEL_MAP[args.model_name]
+        self.comm = Communicator(backend=getattr(args, 'backend', None),
+                                 init_method=getattr(args, 'init_method', 'env://'))
+        self._setup_logging(getattr(args, 'logger_level', 'INFO'))
+        self.comm.sync_seed(getattr(args, 'seed', None) or 42)
+        self._set_graph()
+        self._set_engine()
+        self._set_assigner()
+        self._set_model()
+        self.recorder = Recorder()
+        self.epoch_times = []
+
+    # ------------------------------------------------------------------
+    def _setup_logging(self, level: str):
+        logging.basicConfig(
+            level=getattr(logging, level.upper(), logging.INFO),
+            format=f'[rank {self.comm.rank}] %(asctime)s %(levelname)s %(message)s')
+
+    def _set_graph(self):
+        args = self.args
+        world, rank = self.comm.world_size, self.comm.rank
+        part_dir = getattr(args, 'partition_dir', None) or 'part_data'
+        ds = args.dataset
+        meta = os.path.join(part_dir, ds, f'{world}part', f'{ds}.json')
+        if not os.path.exists(meta):
+            if rank == 0:
+                logger.info('partition cache miss -> generating synthetic '
+                            f'{ds} and partitioning into {world}')
+                dcfg = self.cfg.get('data', {})
+                scale = getattr(args, 'scale', None) or dcfg.get('scale', 1.0)
+                g = synth_graph(ds, world, seed=17,
+                                cut_frac=dcfg.get('cut_frac', 0.1),
+                                scale=scale)
+                save_partitions(partition_all(g, world), part_dir, ds)
+            self.comm.barrier()
+        self.graph = load_partition(part_dir, ds, world, rank)
+
+    def _set_engine(self):
+        m = self.cfg['model']
+        shape = DATASET_SHAPES[self.args.dataset]
+        self.feat_dim, self.num_classes, self.multilabel = shape[2], shape[3], shape[4]
+        L = m['num_layers']
+        msg_dims = [self.feat_dim] + [m['hidden_dim']] * (L - 1)
+        self.engine = GraphEngine(self.graph, self.mode, self.model_type,
+                                  msg_dims, agg_type=m['aggregator_type'],
+                                  device=self.comm.device)
+
+    def _set_assigner(self):
+        a = self.cfg['assignment']
+        scheme = AssignScheme(getattr(self.args, 'assign_scheme', None)
+                              or a['scheme'])
+        self.assigner = Assigner(self.engine, scheme,
+                                 group_size=a['group_size'],
+                                 coe_lambda=a['coe_lambda'],
+                                 init_bits=a['init_bits'],
+                                 profile_data_length=a.get('profile_data_length', 20))
+        self.assign_cycle = a['assign_cycle']
+        if self.mode.bit_type.name == 'QUANT':
+            self.assigner.initial_assignment()
+
+    def _set_model(self):
+        m = self.cfg['model']
+        cls = DistGCN if self.model_type == DistGNNType.DistGCN else DistSAGE
+        kwargs = dict(num_layers=m['num_layers'], dropout=m['dropout'],
+                      use_norm=m['use_norm'])
+        if cls is DistSAGE:
+            kwargs['aggregator_type'] = m['aggregator_type']
+        self.model = cls(self.feat_dim, m['hidden_dim'], self.num_classes,
+                         **kwargs).to(self.comm.device)
+        self.comm.sync_model_params(self.model)
+        self.optimizer = torch.optim.Adam(self.model.parameters(),
+                                          lr=self.cfg['runtime']['lr'])
+
+    # ------------------------------------------------------------------
+    def train(self) -> Dict[str, float]:
+        rt = self.cfg['runtime']
+        gc = global_train_count(self.engine)
+        quant = self.mode.bit_type.name == 'QUANT'
+        adaptive = quant and self.assigner.scheme == AssignScheme.ADAPTIVE
+        for epoch in range(rt['num_epochs']):
+            if adaptive and epoch > 0 and epoch % self.assign_cycle == 0:
+                self.assigner.reassign()
+            t0 = time.perf_counter()
+            loss = train_epoch(self.engine, self.model, self.optimizer, gc,
+                               self.multilabel)
+            if self.engine.device.type == 'cuda':
+                torch.cuda.synchronize()
+            self.epoch_times.append(time.perf_counter() - t0)
+            if epoch % rt.get('eval_every', 1) == 0:
+                metrics = evaluate(self.engine, self.model, self.multilabel)
+                self.recorder.add(metrics)
+            if epoch % rt['log_steps'] == 0 and self.comm.rank == 0:
+                logger.info(
+                    f'epoch {epoch:04d} loss {float(loss):.4f} '
+                    f'train {metrics["train"]:.4f} val {metrics["val"]:.4f} '
+                    f'test {metrics["test"]:.4f} '
+                    f'epoch_time {self.epoch_times[-1]*1e3:.1f}ms')
+        return self.recorder.best()
+
+    def save(self, root: str = 'exp'):
+        args = self.args
+        tag = f'{self.mode.value}'
+        if self.mode.bit_type.name == 'QUANT':
+            tag += f'_{self.assigner.scheme.value}'
+        out = os.path.join(root, args.dataset,
+                           f'{self.comm.world_size}part', args.model_name)
+        if self.comm.rank == 0:
+            times = torch.tensor(self.epoch_times)
+            extra = {
+                'mean_epoch_time_s': f'{times.mean():.4f}' if len(times) else 'n/a',
+                'total_time_s': f'{times.sum():.2f}',
+            }
+            self.recorder.save(os.path.join(out, 'metrics'), tag, extra)
+            os.makedirs(os.path.join(out, 'time'), exist_ok=True)
+            torch.save(times, os.path.join(out, 'time', f'{tag}_epoch_times.pt'))
+            logger.info(f'saved results under {out}')
+        self.comm.barrier()

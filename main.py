@@ -1,0 +1,48 @@
+#!/usr/bin/env python3
+"""Training entry point.
+
+Same CLI surface as the reference (``/root/reference/main.py:6-15``):
+dataset, num_parts (implied by WORLD_SIZE under torchrun), backend,
+init_method, model_name, mode, assign_scheme, logger_level.
+
+Launch (one process per GPU over RCCL):
+    python -m torch.distributed.run --nnodes=1 --nproc-per-node 4 \
+        --master-addr 127.0.0.1 main.py --dataset reddit --model_name gcn \
+        --mode AdaQP
+"""
+import argparse
+
+from adaqp_amd.runtime.trainer import Trainer
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument('--dataset', type=str, default='reddit',
+                   choices=['reddit', 'yelp', 'ogbn-products', 'amazonProducts'])
+    p.add_argument('--model_name', type=str, default='gcn',
+                   choices=['gcn', 'sage'])
+    p.add_argument('--mode', type=str, default='AdaQP',
+                   choices=['Vanilla', 'AdaQP', 'AdaQP-q', 'AdaQP-p'])
+    p.add_argument('--assign_scheme', type=str, default=None,
+                   choices=[None, 'uniform', 'random', 'adaptive'])
+    p.add_argument('--backend', type=str, default=None,
+                   help='torch.distributed backend (default: RCCL on GPU, gloo on CPU)')
+    p.add_argument('--init_method', type=str, default='env://')
+    p.add_argument('--logger_level', type=str, default='INFO')
+    p.add_argument('--partition_dir', type=str, default='part_data')
+    p.add_argument('--num_epochs', type=int, default=None)
+    p.add_argument('--lr', type=float, default=None)
+    p.add_argument('--log_steps', type=int, default=None)
+    p.add_argument('--seed', type=int, default=None)
+    args = p.parse_args()
+
+    trainer = Trainer(args)
+    best = trainer.train()
+    trainer.save()
+    if trainer.comm.rank == 0:
+        print(f'best: epoch {best["epoch"]} val {best["val"]:.4f} '
+              f'test {best["test"]:.4f}')
+
+
+if __name__ == '__main__':
+    main()
