@@ -160,47 +160,82 @@ __global__ void quant_unpack_kernel(
 
 // ---------------------------------------------------------------------------
 // spmm_csr: y[r] = dst_scale[r] * sum_{e in row r} src_scale[c_e] * x[c_e]
-// One wave per row, grid-stride; lane l owns features 4l..4l+3 (float4)
-// and loops over feature chunks of 256.
+//
+// One SUB-wavefront (SW lanes, 16/32/64 picked from F) per destination row —
+// a 64-wide wave covers 64/SW rows so narrow feature dims (ogbn-products
+// F=100 -> SW=32) keep every lane busy. Lane owns features [4*sl, 4*sl+4)
+// (float4) per chunk of 4*SW. Two edges are accumulated in flight (dual
+// accumulators) to cover gather latency. Dual-tensor input: columns
+// >= n_local read the REMOTE block directly (no [N,F] concat per layer).
+// XCD-aware bijective block swizzle gives each XCD a contiguous row chunk
+// so clustered neighbor rows hit the same L2 (guide §5.5 T1).
 // ---------------------------------------------------------------------------
+template <int SW>
 __global__ void spmm_csr_kernel(
     const int64_t* __restrict__ indptr, const int64_t* __restrict__ indices,
-    const float* __restrict__ x, float* __restrict__ y,
+    const float* __restrict__ xl, const float* __restrict__ xr,
+    float* __restrict__ y,
     const float* __restrict__ src_scale, const float* __restrict__ dst_scale,
-    int64_t num_rows, int64_t F) {
-    const int waves_per_block = blockDim.x / WAVE;
-    const int64_t wid0 = (int64_t)blockIdx.x * waves_per_block + threadIdx.x / WAVE;
-    const int64_t stride = (int64_t)gridDim.x * waves_per_block;
-    const int lane = threadIdx.x & (WAVE - 1);
+    int64_t num_rows, int64_t F, int64_t n_local) {
+    const int rows_per_block = blockDim.x / SW;
+    // bijective XCD swizzle: blocks [0,nwg) -> xcd-contiguous chunks
+    const int64_t nwg = gridDim.x;
+    const int64_t q = nwg / 8, rem = nwg % 8;
+    const int64_t xcd = blockIdx.x % 8, idx = blockIdx.x / 8;
+    const int64_t swz = (xcd < rem ? xcd * (q + 1) : rem * (q + 1) + (xcd - rem) * q) + idx;
+    const int64_t sub0 = swz * rows_per_block + threadIdx.x / SW;
+    const int64_t stride = nwg * rows_per_block;
+    const int sl = threadIdx.x & (SW - 1);
 
-    for (int64_t r = wid0; r < num_rows; r += stride) {
+    for (int64_t r = sub0; r < num_rows; r += stride) {
         const int64_t e0 = indptr[r], e1 = indptr[r + 1];
         const float ds = dst_scale ? dst_scale[r] : 1.f;
-        for (int64_t f0 = (int64_t)lane * 4; f0 < F; f0 += WAVE * 4) {
-            float4 acc = {0.f, 0.f, 0.f, 0.f};
+        for (int64_t f0 = (int64_t)sl * 4; f0 < F; f0 += (int64_t)SW * 4) {
+            float4 acc0 = {0.f, 0.f, 0.f, 0.f};
+            float4 acc1 = {0.f, 0.f, 0.f, 0.f};
             const bool full = (f0 + 4 <= F);
-            for (int64_t e = e0; e < e1; ++e) {
+            int64_t e = e0;
+            if (full) {
+                for (; e + 1 < e1; e += 2) {
+                    const int64_t c0 = indices[e], c1 = indices[e + 1];
+                    const float s0 = src_scale ? src_scale[c0] : 1.f;
+                    const float s1 = src_scale ? src_scale[c1] : 1.f;
+                    const float* p0 = (c0 < n_local ? xl + c0 * F
+                                                    : xr + (c0 - n_local) * F) + f0;
+                    const float* p1 = (c1 < n_local ? xl + c1 * F
+                                                    : xr + (c1 - n_local) * F) + f0;
+                    const float4 v0 = *reinterpret_cast<const float4*>(p0);
+                    const float4 v1 = *reinterpret_cast<const float4*>(p1);
+                    acc0.x = fmaf(v0.x, s0, acc0.x); acc1.x = fmaf(v1.x, s1, acc1.x);
+                    acc0.y = fmaf(v0.y, s0, acc0.y); acc1.y = fmaf(v1.y, s1, acc1.y);
+                    acc0.z = fmaf(v0.z, s0, acc0.z); acc1.z = fmaf(v1.z, s1, acc1.z);
+                    acc0.w = fmaf(v0.w, s0, acc0.w); acc1.w = fmaf(v1.w, s1, acc1.w);
+                }
+            }
+            for (; e < e1; ++e) {
                 const int64_t c = indices[e];
                 const float s = src_scale ? src_scale[c] : 1.f;
-                const float* xc = x + c * F + f0;
+                const float* xc = (c < n_local ? xl + c * F
+                                               : xr + (c - n_local) * F) + f0;
                 if (full) {
                     const float4 v = *reinterpret_cast<const float4*>(xc);
-                    acc.x = fmaf(v.x, s, acc.x);
-                    acc.y = fmaf(v.y, s, acc.y);
-                    acc.z = fmaf(v.z, s, acc.z);
-                    acc.w = fmaf(v.w, s, acc.w);
+                    acc0.x = fmaf(v.x, s, acc0.x);
+                    acc0.y = fmaf(v.y, s, acc0.y);
+                    acc0.z = fmaf(v.z, s, acc0.z);
+                    acc0.w = fmaf(v.w, s, acc0.w);
                 } else {
                     for (int k = 0; k < 4 && f0 + k < F; ++k)
-                        ((float*)&acc)[k] = fmaf(xc[k], s, ((float*)&acc)[k]);
+                        ((float*)&acc0)[k] = fmaf(xc[k], s, ((float*)&acc0)[k]);
                 }
             }
             float* yr = y + r * F + f0;
             if (full) {
-                acc.x *= ds; acc.y *= ds; acc.z *= ds; acc.w *= ds;
-                *reinterpret_cast<float4*>(yr) = acc;
+                const float4 o = {(acc0.x + acc1.x) * ds, (acc0.y + acc1.y) * ds,
+                                  (acc0.z + acc1.z) * ds, (acc0.w + acc1.w) * ds};
+                *reinterpret_cast<float4*>(yr) = o;
             } else {
                 for (int k = 0; k < 4 && f0 + k < F; ++k)
-                    yr[k] = ((float*)&acc)[k] * ds;
+                    yr[k] = (((float*)&acc0)[k] + ((float*)&acc1)[k]) * ds;
             }
         }
     }
@@ -263,24 +298,40 @@ void quant_unpack(torch::Tensor payload, torch::Tensor params, torch::Tensor row
 #undef LAUNCH
 }
 
-void spmm_csr(torch::Tensor indptr, torch::Tensor indices, torch::Tensor x,
-              torch::Tensor y, torch::Tensor src_scale, torch::Tensor dst_scale) {
-    CHECK_DEV(x); CHECK_CONTIG(x); CHECK_DEV(y); CHECK_CONTIG(y);
-    TORCH_CHECK(x.scalar_type() == torch::kFloat32, "spmm_csr expects fp32");
-    const int64_t num_rows = y.size(0), F = x.size(1);
+void spmm_csr(torch::Tensor indptr, torch::Tensor indices, torch::Tensor xl,
+              torch::Tensor xr, torch::Tensor y, torch::Tensor src_scale,
+              torch::Tensor dst_scale) {
+    CHECK_DEV(xl); CHECK_CONTIG(xl); CHECK_DEV(y); CHECK_CONTIG(y);
+    TORCH_CHECK(xl.scalar_type() == torch::kFloat32, "spmm_csr expects fp32");
+    const int64_t num_rows = y.size(0), F = xl.size(1);
+    const int64_t n_local = xl.size(0);
     TORCH_CHECK(indptr.numel() == num_rows + 1, "indptr/num_rows mismatch");
-    const int waves_per_block = 4;
-    const dim3 block(WAVE * waves_per_block);
-    // >> 256 workgroups to fill 8 XCDs; cap and grid-stride
-    int64_t blocks = (num_rows + waves_per_block - 1) / waves_per_block;
-    blocks = std::min<int64_t>(blocks, 8192);
-    const dim3 grid(std::max<int64_t>(blocks, 1));
-    spmm_csr_kernel<<<grid, block, 0, cur_stream()>>>(
-        indptr.data_ptr<int64_t>(), indices.data_ptr<int64_t>(),
-        x.data_ptr<float>(), y.data_ptr<float>(),
-        src_scale.numel() ? src_scale.data_ptr<float>() : nullptr,
-        dst_scale.numel() ? dst_scale.data_ptr<float>() : nullptr,
-        num_rows, F);
+    const float* xr_ptr = nullptr;
+    if (xr.numel()) {
+        CHECK_DEV(xr); CHECK_CONTIG(xr);
+        TORCH_CHECK(xr.size(1) == F, "remote feature dim mismatch");
+        xr_ptr = xr.data_ptr<float>();
+    }
+    // sub-wavefront width from F: keep all 64 lanes of a wave busy
+    const int sw = F > 128 ? 64 : (F > 64 ? 32 : 16);
+    const int block_threads = WAVE * 4;
+    const int rows_per_block = block_threads / sw;
+    int64_t blocks = (num_rows + rows_per_block - 1) / rows_per_block;
+    blocks = std::max<int64_t>(std::min<int64_t>(blocks, 16384), 1);
+    const dim3 grid(blocks), block(block_threads);
+    auto s = cur_stream();
+#define SPMM_LAUNCH(SW) spmm_csr_kernel<SW><<<grid, block, 0, s>>>( \
+        indptr.data_ptr<int64_t>(), indices.data_ptr<int64_t>(), \
+        xl.data_ptr<float>(), xr_ptr, y.data_ptr<float>(), \
+        src_scale.numel() ? src_scale.data_ptr<float>() : nullptr, \
+        dst_scale.numel() ? dst_scale.data_ptr<float>() : nullptr, \
+        num_rows, F, n_local)
+    switch (sw) {
+        case 64: SPMM_LAUNCH(64); break;
+        case 32: SPMM_LAUNCH(32); break;
+        default: SPMM_LAUNCH(16); break;
+    }
+#undef SPMM_LAUNCH
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

@@ -95,14 +95,13 @@ def _scales(engine, mode: PropagationMode) -> Tuple[Optional[Tensor], Optional[T
     raise ValueError(f'unknown aggregator {engine.agg_type}')
 
 
-def _agg(engine, view, x_full: Tensor, src_scale, dst_scale, row_lo, row_hi) -> Tensor:
-    """SpMM over rows [row_lo, row_hi) of a (indptr, indices, base, nrows) view."""
+def _agg(engine, view, x_local: Tensor, x_remote, src_scale, dst_scale) -> Tensor:
+    """SpMM over one (indptr, indices, base, nrows) view. src_scale is the
+    full [N] vector; dst_scale is sliced to the view's row range."""
     indptr, indices, base, nrows = view
-    y = spmm(indptr, indices, x_full,
-             src_scale[:x_full.shape[0]] if src_scale is not None else None,
-             dst_scale[base:base + nrows] if dst_scale is not None else None,
-             nrows)
-    return y
+    return spmm(indptr, indices, x_local, x_remote, src_scale,
+                dst_scale[base:base + nrows] if dst_scale is not None else None,
+                nrows)
 
 
 def _self_term(engine, x_local: Tensor, mode: PropagationMode) -> Tensor:
@@ -120,11 +119,9 @@ def _self_term(engine, x_local: Tensor, mode: PropagationMode) -> Tensor:
 def full_propagation(engine, x_local: Tensor, key: str, is_train: bool,
                      mode: PropagationMode) -> Tensor:
     remote = _exchange(engine, x_local, key, is_train)
-    x_full = torch.cat([x_local, remote], dim=0)
     src_scale, dst_scale, add_self = _scales(engine, mode)
     with engine.timer.record(f'{key}_full_aggregation'):
-        y = _agg(engine, engine.full_view, x_full, src_scale, dst_scale,
-                 0, engine.graph.num_inner)
+        y = _agg(engine, engine.full_view, x_local, remote, src_scale, dst_scale)
         if add_self:
             y = y + _self_term(engine, x_local, mode)
     return y
@@ -153,17 +150,16 @@ def decomposed_propagation(engine, x_local: Tensor, key: str, is_train: bool,
     with ctx:
         # central rows only touch local columns -> safe to run now
         with engine.timer.record(f'{key}_central_aggregation'):
-            y_c = _agg(engine, engine.central_view, x_local, src_scale,
-                       dst_scale, 0, g.num_central)
+            y_c = _agg(engine, engine.central_view, x_local, None, src_scale,
+                       dst_scale)
 
     if on_gpu:
         torch.cuda.current_stream().wait_event(engine.remote_ready)
         remote.record_stream(torch.cuda.current_stream())
 
-    x_full = torch.cat([x_local, remote], dim=0)
     with engine.timer.record(f'{key}_marginal_aggregation'):
-        y_m = _agg(engine, engine.marginal_view, x_full, src_scale,
-                   dst_scale, g.num_central, g.num_marginal)
+        y_m = _agg(engine, engine.marginal_view, x_local, remote, src_scale,
+                   dst_scale)
     y = torch.cat([y_c, y_m], dim=0)
     if add_self:
         y = y + _self_term(engine, x_local, mode)

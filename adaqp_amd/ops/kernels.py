@@ -107,21 +107,28 @@ def mixed_dequantize(payload: Tensor, params: Tensor, plan: SidePlan,
 _csr_cache: dict = {}
 
 
-def spmm(indptr: Tensor, indices: Tensor, x: Tensor,
+def spmm(indptr: Tensor, indices: Tensor, x_local: Tensor,
+         x_remote: Optional[Tensor],
          src_scale: Optional[Tensor], dst_scale: Optional[Tensor],
          num_rows: int) -> Tensor:
     """Aggregation SpMM over an in-edge CSR.
 
-    x: [N, F] full feature matrix (local rows first, then remote).
-    src_scale: [N] or None; dst_scale: [num_rows] or None.
+    Columns < len(x_local) read x_local; the rest read x_remote (the
+    all-to-all output block) — no concat on the GPU path.
+    src_scale: [n_local+n_remote] or None; dst_scale: [num_rows] or None.
     """
-    if x.is_cuda:
-        out = torch.empty(num_rows, x.shape[1], dtype=x.dtype, device=x.device)
-        native().spmm_csr(indptr, indices, x, out,
-                          src_scale if src_scale is not None else torch.empty(0, device=x.device),
-                          dst_scale if dst_scale is not None else torch.empty(0, device=x.device))
+    if x_local.is_cuda:
+        out = torch.empty(num_rows, x_local.shape[1], dtype=x_local.dtype,
+                          device=x_local.device)
+        empty = torch.empty(0, device=x_local.device)
+        native().spmm_csr(indptr, indices, x_local,
+                          x_remote if x_remote is not None else empty, out,
+                          src_scale if src_scale is not None else empty,
+                          dst_scale if dst_scale is not None else empty)
         return out
-    xs = x * src_scale[:, None] if src_scale is not None else x
+    x = (torch.cat([x_local, x_remote], dim=0)
+         if x_remote is not None and x_remote.numel() else x_local)
+    xs = x * src_scale[:x.shape[0], None] if src_scale is not None else x
     key = (indptr.data_ptr(), indices.data_ptr(), num_rows, x.shape[0])
     sp = _csr_cache.get(key)
     if sp is None:

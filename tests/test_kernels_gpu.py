@@ -101,7 +101,10 @@ def test_spmm_matches_torch(F):
     src = torch.rand(lg.num_nodes, device='cuda') + 0.5
     dst = torch.rand(lg.num_inner, device='cuda') + 0.5
     y = torch.empty(lg.num_inner, F, device='cuda')
-    C.spmm_csr(lg.indptr.cuda(), lg.indices.cuda(), x, y, src, dst)
+    # exercise the dual-tensor path: split x arbitrarily at num_inner
+    xl = x[:lg.num_inner].contiguous()
+    xr = x[lg.num_inner:].contiguous()
+    C.spmm_csr(lg.indptr.cuda(), lg.indices.cuda(), xl, xr, y, src, dst)
     # torch fp32 reference
     sp = torch.sparse_csr_tensor(lg.indptr, lg.indices,
                                  torch.ones(lg.num_edges),
@@ -119,7 +122,7 @@ def test_spmm_empty_rows():
     x = torch.randn(4, 8, device='cuda')
     y = torch.empty(3, 8, device='cuda')
     empty = torch.empty(0, device='cuda')
-    C.spmm_csr(indptr, indices, x, y, empty, empty)
+    C.spmm_csr(indptr, indices, x, empty, y, empty, empty)
     assert torch.allclose(y[0], torch.zeros(8, device='cuda'))
     assert torch.allclose(y[1].cpu(), x[0].cpu() + x[2].cpu(), atol=1e-5)
     assert torch.allclose(y[2], torch.zeros(8, device='cuda'))
