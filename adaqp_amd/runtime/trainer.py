@@ -101,6 +101,8 @@ class Trainer:
         self.engine = GraphEngine(self.graph, self.mode, self.model_type,
                                   msg_dims, agg_type=m['aggregator_type'],
                                   device=self.comm.device)
+        if getattr(self.args, 'time_breakdown', False):
+            self.engine.timer.enabled = True
 
     def _set_assigner(self):
         a = self.cfg['assignment']
@@ -143,6 +145,8 @@ class Trainer:
             if self.engine.device.type == 'cuda':
                 torch.cuda.synchronize()
             self.epoch_times.append(time.perf_counter() - t0)
+            if self.engine.timer.enabled:
+                self.engine.timer.epoch_rollup()
             if epoch % rt.get('eval_every', 1) == 0:
                 metrics = evaluate(self.engine, self.model, self.multilabel)
                 self.recorder.add(metrics)
@@ -170,5 +174,11 @@ class Trainer:
             self.recorder.save(os.path.join(out, 'metrics'), tag, extra)
             os.makedirs(os.path.join(out, 'time'), exist_ok=True)
             torch.save(times, os.path.join(out, 'time', f'{tag}_epoch_times.pt'))
+            if self.engine.timer.epoch_rows:
+                # per-epoch span breakdown CSV (reference trainer.py:229)
+                with open(os.path.join(out, 'time', f'{tag}_breakdown.csv'), 'w') as f:
+                    f.write('comm,quant,central,marginal,full,reduce\n')
+                    for row in self.engine.timer.epoch_rows:
+                        f.write(','.join(f'{v:.6f}' for v in row) + '\n')
             logger.info(f'saved results under {out}')
         self.comm.barrier()

@@ -34,7 +34,18 @@ def main():
     p.add_argument('--lr', type=float, default=None)
     p.add_argument('--log_steps', type=int, default=None)
     p.add_argument('--seed', type=int, default=None)
+    p.add_argument('--time_breakdown', action='store_true',
+                   help='enable per-epoch comm/quant/agg span timing '
+                        '(adds sync fences; use rocprofv3 for kernel evidence)')
     args = p.parse_args()
+
+    # allow plain `python main.py` without torchrun (world_size 1)
+    import os
+    os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
+    os.environ.setdefault('MASTER_PORT', '29501')
+    os.environ.setdefault('RANK', '0')
+    os.environ.setdefault('WORLD_SIZE', '1')
+    os.environ.setdefault('LOCAL_RANK', os.environ.get('RANK', '0'))
 
     trainer = Trainer(args)
     best = trainer.train()
