@@ -170,13 +170,26 @@ __global__ void quant_unpack_kernel(
 // XCD-aware bijective block swizzle gives each XCD a contiguous row chunk
 // so clustered neighbor rows hit the same L2 (guide §5.5 T1).
 // ---------------------------------------------------------------------------
+__global__ void zero_rows_kernel(float* __restrict__ y,
+                                 const int32_t* __restrict__ rows,
+                                 int64_t n, int64_t F) {
+    const int64_t wid = (int64_t)blockIdx.x * (blockDim.x / WAVE)
+                      + threadIdx.x / WAVE;
+    if (wid >= n) return;
+    float* p = y + (int64_t)rows[wid] * F;
+    for (int64_t f = threadIdx.x & (WAVE - 1); f < F; f += WAVE)
+        p[f] = 0.f;
+}
+
 template <int SW>
 __global__ void spmm_csr_kernel(
-    const int64_t* __restrict__ indptr, const int64_t* __restrict__ indices,
+    const int64_t* __restrict__ indices,
     const float* __restrict__ xl, const float* __restrict__ xr,
     float* __restrict__ y,
     const float* __restrict__ src_scale, const float* __restrict__ dst_scale,
-    int64_t num_rows, int64_t F, int64_t n_local) {
+    const int32_t* __restrict__ seg_row, const int64_t* __restrict__ seg_e0,
+    const int64_t* __restrict__ seg_e1, const uint8_t* __restrict__ seg_multi,
+    int64_t n_seg, int64_t F, int64_t n_local) {
     const int rows_per_block = blockDim.x / SW;
     // bijective XCD swizzle: blocks [0,nwg) -> xcd-contiguous chunks
     const int64_t nwg = gridDim.x;
@@ -187,8 +200,10 @@ __global__ void spmm_csr_kernel(
     const int64_t stride = nwg * rows_per_block;
     const int sl = threadIdx.x & (SW - 1);
 
-    for (int64_t r = sub0; r < num_rows; r += stride) {
-        const int64_t e0 = indptr[r], e1 = indptr[r + 1];
+    for (int64_t it = sub0; it < n_seg; it += stride) {
+        const int64_t r = seg_row[it];
+        const int64_t e0 = seg_e0[it], e1 = seg_e1[it];
+        const bool multi = seg_multi[it];
         const float ds = dst_scale ? dst_scale[r] : 1.f;
         for (int64_t f0 = (int64_t)sl * 4; f0 < F; f0 += (int64_t)SW * 4) {
             float4 acc0 = {0.f, 0.f, 0.f, 0.f};
@@ -229,13 +244,17 @@ __global__ void spmm_csr_kernel(
                 }
             }
             float* yr = y + r * F + f0;
-            if (full) {
+            if (full && !multi) {
                 const float4 o = {(acc0.x + acc1.x) * ds, (acc0.y + acc1.y) * ds,
                                   (acc0.z + acc1.z) * ds, (acc0.w + acc1.w) * ds};
                 *reinterpret_cast<float4*>(yr) = o;
-            } else {
+            } else if (!multi) {
                 for (int k = 0; k < 4 && f0 + k < F; ++k)
                     yr[k] = (((float*)&acc0)[k] + ((float*)&acc1)[k]) * ds;
+            } else {
+                // long row split across segments: combine via atomics
+                for (int k = 0; k < 4 && f0 + k < F; ++k)
+                    atomicAdd(yr + k, (((float*)&acc0)[k] + ((float*)&acc1)[k]) * ds);
             }
         }
     }
@@ -298,34 +317,45 @@ void quant_unpack(torch::Tensor payload, torch::Tensor params, torch::Tensor row
 #undef LAUNCH
 }
 
-void spmm_csr(torch::Tensor indptr, torch::Tensor indices, torch::Tensor xl,
+void spmm_csr(torch::Tensor indices, torch::Tensor xl,
               torch::Tensor xr, torch::Tensor y, torch::Tensor src_scale,
-              torch::Tensor dst_scale) {
+              torch::Tensor dst_scale, torch::Tensor seg_row,
+              torch::Tensor seg_e0, torch::Tensor seg_e1,
+              torch::Tensor seg_multi, torch::Tensor zero_rows) {
     CHECK_DEV(xl); CHECK_CONTIG(xl); CHECK_DEV(y); CHECK_CONTIG(y);
     TORCH_CHECK(xl.scalar_type() == torch::kFloat32, "spmm_csr expects fp32");
-    const int64_t num_rows = y.size(0), F = xl.size(1);
+    const int64_t F = xl.size(1);
     const int64_t n_local = xl.size(0);
-    TORCH_CHECK(indptr.numel() == num_rows + 1, "indptr/num_rows mismatch");
+    const int64_t n_seg = seg_row.numel();
     const float* xr_ptr = nullptr;
     if (xr.numel()) {
         CHECK_DEV(xr); CHECK_CONTIG(xr);
         TORCH_CHECK(xr.size(1) == F, "remote feature dim mismatch");
         xr_ptr = xr.data_ptr<float>();
     }
+    auto s = cur_stream();
+    if (zero_rows.numel()) {
+        const dim3 zb(256), zg((zero_rows.numel() + 3) / 4);
+        zero_rows_kernel<<<zg, zb, 0, s>>>(y.data_ptr<float>(),
+                                           zero_rows.data_ptr<int32_t>(),
+                                           zero_rows.numel(), F);
+    }
+    if (n_seg == 0) return;
     // sub-wavefront width from F: keep all 64 lanes of a wave busy
     const int sw = F > 128 ? 64 : (F > 64 ? 32 : 16);
     const int block_threads = WAVE * 4;
     const int rows_per_block = block_threads / sw;
-    int64_t blocks = (num_rows + rows_per_block - 1) / rows_per_block;
+    int64_t blocks = (n_seg + rows_per_block - 1) / rows_per_block;
     blocks = std::max<int64_t>(std::min<int64_t>(blocks, 16384), 1);
     const dim3 grid(blocks), block(block_threads);
-    auto s = cur_stream();
 #define SPMM_LAUNCH(SW) spmm_csr_kernel<SW><<<grid, block, 0, s>>>( \
-        indptr.data_ptr<int64_t>(), indices.data_ptr<int64_t>(), \
+        indices.data_ptr<int64_t>(), \
         xl.data_ptr<float>(), xr_ptr, y.data_ptr<float>(), \
         src_scale.numel() ? src_scale.data_ptr<float>() : nullptr, \
         dst_scale.numel() ? dst_scale.data_ptr<float>() : nullptr, \
-        num_rows, F, n_local)
+        seg_row.data_ptr<int32_t>(), seg_e0.data_ptr<int64_t>(), \
+        seg_e1.data_ptr<int64_t>(), seg_multi.data_ptr<uint8_t>(), \
+        n_seg, F, n_local)
     switch (sw) {
         case 64: SPMM_LAUNCH(64); break;
         case 32: SPMM_LAUNCH(32); break;

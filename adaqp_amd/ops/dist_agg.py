@@ -96,12 +96,11 @@ def _scales(engine, mode: PropagationMode) -> Tuple[Optional[Tensor], Optional[T
 
 
 def _agg(engine, view, x_local: Tensor, x_remote, src_scale, dst_scale) -> Tensor:
-    """SpMM over one (indptr, indices, base, nrows) view. src_scale is the
-    full [N] vector; dst_scale is sliced to the view's row range."""
-    indptr, indices, base, nrows = view
-    return spmm(indptr, indices, x_local, x_remote, src_scale,
-                dst_scale[base:base + nrows] if dst_scale is not None else None,
-                nrows)
+    """SpMM over one SpmmView. src_scale is the full [N] vector; dst_scale
+    is sliced to the view's row range."""
+    ds = (dst_scale[view.base:view.base + view.nrows]
+          if dst_scale is not None else None)
+    return spmm(view, x_local, x_remote, src_scale, ds)
 
 
 def _self_term(engine, x_local: Tensor, mode: PropagationMode) -> Tensor:

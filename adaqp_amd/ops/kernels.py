@@ -106,25 +106,64 @@ def mixed_dequantize(payload: Tensor, params: Tensor, plan: SidePlan,
 
 _csr_cache: dict = {}
 
+SEG_EDGES = 512      # max edges one sub-wavefront processes serially
 
-def spmm(indptr: Tensor, indices: Tensor, x_local: Tensor,
-         x_remote: Optional[Tensor],
-         src_scale: Optional[Tensor], dst_scale: Optional[Tensor],
-         num_rows: int) -> Tensor:
-    """Aggregation SpMM over an in-edge CSR.
+
+class SpmmView:
+    """One CSR row range + its work-item segmentation.
+
+    Power-law graphs put 10^4-10^5 edges on hub rows; a single
+    sub-wavefront grinding one such row serially becomes the kernel's
+    critical path. Rows are split into <=SEG_EDGES-edge segments; rows
+    with >1 segment combine via global atomics into pre-zeroed output
+    rows (SURVEY.md §7 'hard parts': row-binning / load imbalance).
+    """
+
+    def __init__(self, indptr: Tensor, indices: Tensor, base: int, nrows: int):
+        self.indptr = indptr
+        self.indices = indices
+        self.base = int(base)
+        self.nrows = int(nrows)
+        counts = (indptr[1:] - indptr[:-1]).cpu()
+        nseg = torch.clamp((counts + SEG_EDGES - 1) // SEG_EDGES, min=1)
+        seg_row = torch.repeat_interleave(torch.arange(nrows, dtype=torch.int64), nseg)
+        first = torch.repeat_interleave(torch.cumsum(nseg, 0) - nseg, nseg)
+        seg_in_row = torch.arange(seg_row.numel(), dtype=torch.int64) - first
+        e0 = indptr.cpu()[seg_row] + seg_in_row * SEG_EDGES
+        e1 = torch.minimum(e0 + SEG_EDGES, indptr.cpu()[seg_row + 1])
+        multi_mask = nseg > 1
+        self.seg_row = seg_row.to(torch.int32)
+        self.seg_e0 = e0
+        self.seg_e1 = e1
+        self.seg_multi = multi_mask[seg_row].to(torch.uint8)
+        self.zero_rows = torch.nonzero(multi_mask, as_tuple=True)[0].to(torch.int32)
+        self._on = None
+
+    def to(self, device):
+        for n in ('seg_row', 'seg_e0', 'seg_e1', 'seg_multi', 'zero_rows'):
+            setattr(self, n, getattr(self, n).to(device))
+        return self
+
+
+def spmm(view: 'SpmmView', x_local: Tensor, x_remote: Optional[Tensor],
+         src_scale: Optional[Tensor], dst_scale: Optional[Tensor]) -> Tensor:
+    """Aggregation SpMM over an in-edge CSR view.
 
     Columns < len(x_local) read x_local; the rest read x_remote (the
     all-to-all output block) — no concat on the GPU path.
-    src_scale: [n_local+n_remote] or None; dst_scale: [num_rows] or None.
+    src_scale: [n_local+n_remote] or None; dst_scale: [nrows] or None.
     """
+    indptr, indices, num_rows = view.indptr, view.indices, view.nrows
     if x_local.is_cuda:
         out = torch.empty(num_rows, x_local.shape[1], dtype=x_local.dtype,
                           device=x_local.device)
         empty = torch.empty(0, device=x_local.device)
-        native().spmm_csr(indptr, indices, x_local,
+        native().spmm_csr(indices, x_local,
                           x_remote if x_remote is not None else empty, out,
                           src_scale if src_scale is not None else empty,
-                          dst_scale if dst_scale is not None else empty)
+                          dst_scale if dst_scale is not None else empty,
+                          view.seg_row, view.seg_e0, view.seg_e1,
+                          view.seg_multi, view.zero_rows)
         return out
     x = (torch.cat([x_local, x_remote], dim=0)
          if x_remote is not None and x_remote.numel() else x_local)

@@ -69,12 +69,16 @@ class GraphEngine:
         self.sage1_dst_f = (din + 1.0).pow(-1.0)[:I]
         self.sage1_src_b = (din + 1.0).pow(-1.0)
 
-        # decomposition views (zero-copy)
-        self.full_view = (g.indptr, g.indices, 0, g.num_inner)
+        # decomposition views (zero-copy row-range splits + segmentation)
+        from ..ops.kernels import SpmmView
+        self.full_view = SpmmView(g.indptr, g.indices, 0, g.num_inner)
         cptr, cidx, _ = g.central_view()
-        self.central_view = (cptr, cidx, 0, g.num_central)
+        self.central_view = SpmmView(cptr, cidx, 0, g.num_central)
         mptr, midx, mbase = g.marginal_view()
-        self.marginal_view = (mptr, midx, mbase, g.num_marginal)
+        self.marginal_view = SpmmView(mptr, midx, mbase, g.num_marginal)
+        if self.device.type == 'cuda':
+            for v in (self.full_view, self.central_view, self.marginal_view):
+                v.to(self.device)
 
         # overlap streams/events
         if self.device.type == 'cuda':

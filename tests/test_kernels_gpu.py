@@ -93,19 +93,19 @@ def test_quant_unbiased_gpu():
 @pytest.mark.parametrize('F', [64, 100, 256])
 def test_spmm_matches_torch(F):
     from adaqp_amd.graph import random_partitioned_graph, partition_all
-    C = _native()
+    from adaqp_amd.ops.kernels import SpmmView, spmm
+    _native()
     torch.manual_seed(3)
     g = random_partitioned_graph(500, 4000, 4, 3, 1, seed=9)
     lg = partition_all(g, 1)[0]
     x = torch.randn(lg.num_nodes, F, device='cuda')
     src = torch.rand(lg.num_nodes, device='cuda') + 0.5
     dst = torch.rand(lg.num_inner, device='cuda') + 0.5
-    y = torch.empty(lg.num_inner, F, device='cuda')
     # exercise the dual-tensor path: split x arbitrarily at num_inner
     xl = x[:lg.num_inner].contiguous()
     xr = x[lg.num_inner:].contiguous()
-    C.spmm_csr(lg.indptr.cuda(), lg.indices.cuda(), xl, xr, y, src, dst)
-    # torch fp32 reference
+    view = SpmmView(lg.indptr.cuda(), lg.indices.cuda(), 0, lg.num_inner).to('cuda')
+    y = spmm(view, xl, xr, src, dst)
     sp = torch.sparse_csr_tensor(lg.indptr, lg.indices,
                                  torch.ones(lg.num_edges),
                                  size=(lg.num_inner, lg.num_nodes))
@@ -115,14 +115,40 @@ def test_spmm_matches_torch(F):
         (y.cpu() - ref).abs().max()
 
 
+def test_spmm_long_row_segmentation():
+    """A hub row longer than SEG_EDGES must be split and atomically
+    combined; result must still match the dense reference."""
+    from adaqp_amd.ops.kernels import SpmmView, spmm, SEG_EDGES
+    _native()
+    torch.manual_seed(4)
+    N, F = 300, 128
+    hub_deg = 3 * SEG_EDGES + 17
+    rows = [torch.randint(0, N, (hub_deg,))] + \
+           [torch.randint(0, N, (torch.randint(1, 9, (1,)).item(),))
+            for _ in range(49)]
+    indptr = torch.zeros(51, dtype=torch.int64)
+    indptr[1:] = torch.cumsum(torch.tensor([r.numel() for r in rows]), 0)
+    indices = torch.cat(rows)
+    x = torch.randn(N, F, device='cuda')
+    view = SpmmView(indptr.cuda(), indices.cuda(), 0, 50).to('cuda')
+    assert view.zero_rows.numel() >= 1
+    y = spmm(view, x, None, None, None)
+    A = torch.zeros(50, N)
+    for r in range(50):
+        for c in indices[indptr[r]:indptr[r + 1]]:
+            A[r, c] += 1
+    ref = A @ x.cpu()
+    assert torch.allclose(y.cpu(), ref, atol=1e-2, rtol=1e-4)
+
+
 def test_spmm_empty_rows():
-    C = _native()
+    from adaqp_amd.ops.kernels import SpmmView, spmm
+    _native()
     indptr = torch.tensor([0, 0, 2, 2], dtype=torch.int64, device='cuda')
     indices = torch.tensor([0, 2], dtype=torch.int64, device='cuda')
     x = torch.randn(4, 8, device='cuda')
-    y = torch.empty(3, 8, device='cuda')
-    empty = torch.empty(0, device='cuda')
-    C.spmm_csr(indptr, indices, x, empty, y, empty, empty)
+    view = SpmmView(indptr, indices, 0, 3).to('cuda')
+    y = spmm(view, x, None, None, None)
     assert torch.allclose(y[0], torch.zeros(8, device='cuda'))
     assert torch.allclose(y[1].cpu(), x[0].cpu() + x[2].cpu(), atol=1e-5)
     assert torch.allclose(y[2], torch.zeros(8, device='cuda'))
