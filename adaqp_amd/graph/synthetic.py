@@ -59,7 +59,8 @@ def _dedup_edges(src: Tensor, dst: Tensor, n: int) -> Tuple[Tensor, Tensor]:
 
 
 def synth_graph(name: str, num_parts: int, seed: int = 0, cut_frac: float = 0.10,
-                scale: float = 1.0, feat_dim: Optional[int] = None) -> GlobalGraph:
+                scale: float = 1.0, feat_dim: Optional[int] = None,
+                teacher_labels: bool = True) -> GlobalGraph:
     """Build a synthetic graph shaped like dataset ``name``.
 
     ``scale`` < 1 shrinks nodes+edges proportionally (for quick tests).
@@ -72,13 +73,32 @@ def synth_graph(name: str, num_parts: int, seed: int = 0, cut_frac: float = 0.10
     if feat_dim is not None:
         f = feat_dim
     return random_partitioned_graph(n, e, f, c, num_parts, seed=seed,
-                                    cut_frac=cut_frac, multilabel=multilabel)
+                                    cut_frac=cut_frac, multilabel=multilabel,
+                                    teacher_labels=teacher_labels)
+
+
+def _teacher_labels(src: Tensor, dst: Tensor, feats: Tensor, num_classes: int,
+                    gen: torch.Generator) -> Tensor:
+    """Labels from a random 1-hop mean-aggregation linear teacher, so synthetic
+    graphs are LEARNABLE (accuracy comparisons Vanilla vs AdaQP are
+    meaningful — the reference uses real labeled datasets)."""
+    n, f = feats.shape
+    dev = 'cuda' if torch.cuda.is_available() else 'cpu'
+    fd = feats.to(dev)
+    sd, dd = src.to(dev), dst.to(dev)
+    deg = torch.bincount(dd, minlength=n).float().clamp(min=1)
+    agg = torch.zeros_like(fd)
+    agg.index_add_(0, dd, fd[sd])
+    h = agg / deg[:, None]
+    wout = torch.randn(f, num_classes, generator=gen).to(dev)
+    return (h @ wout).argmax(dim=1).cpu()
 
 
 def random_partitioned_graph(num_nodes: int, num_edges: int, feat_dim: int,
                              num_classes: int, num_parts: int, *, seed: int = 0,
                              cut_frac: float = 0.10, multilabel: bool = False,
-                             alpha: float = 2.0) -> GlobalGraph:
+                             alpha: float = 2.0,
+                             teacher_labels: bool = False) -> GlobalGraph:
     g = torch.Generator().manual_seed(seed)
     P = max(num_parts, 1)
     bounds = torch.linspace(0, num_nodes, P + 1, dtype=torch.int64)
@@ -109,6 +129,8 @@ def random_partitioned_graph(num_nodes: int, num_edges: int, feat_dim: int,
     feats = torch.randn(num_nodes, feat_dim, generator=g)
     if multilabel:
         labels = (torch.rand(num_nodes, num_classes, generator=g) < 0.1).float()
+    elif teacher_labels:
+        labels = _teacher_labels(s, d, feats, num_classes, g)
     else:
         labels = torch.randint(0, num_classes, (num_nodes,), generator=g)
     r = torch.rand(num_nodes, generator=g)
