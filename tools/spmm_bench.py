@@ -3,8 +3,11 @@
 rocprofv3). Builds a products-like CSR directly on the GPU and times the
 spmm_csr kernel; prints effective gather TB/s."""
 import argparse
+import os
 import sys
 import time
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), '..'))
 
 import torch
 
