@@ -1,0 +1,67 @@
+"""FastLinear: x @ W + b with a GEMM-based bias gradient.
+
+torch's autograd computes grad_bias as ``grad.sum(dim=0)``, which on
+ROCm lowers to a strided column reduction running at ~0.2 TB/s for the
+[N, 256] grads this model produces (measured: 3.8 ms for 0.73M x 256 —
+~10% of an epoch). Computing it as ``ones[1,N] @ grad`` instead routes
+it through rocBLAS at memory bandwidth.
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+from torch import Tensor
+
+_ones_cache: dict = {}
+
+
+def _ones_row(n: int, device, dtype) -> Tensor:
+    key = (n, device, dtype)
+    t = _ones_cache.get(key)
+    if t is None or t.shape[1] < n:
+        t = torch.ones(1, n, device=device, dtype=dtype)
+        if len(_ones_cache) > 16:
+            _ones_cache.clear()
+        _ones_cache[key] = t
+    return t[:, :n]
+
+
+class _LinearBiasFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: Tensor, w: Tensor, b: Tensor) -> Tensor:
+        ctx.save_for_backward(x, w)
+        return torch.addmm(b, x, w)
+
+    @staticmethod
+    def backward(ctx, g: Tensor):
+        x, w = ctx.saved_tensors
+        g = g.contiguous()
+        gx = g @ w.t() if ctx.needs_input_grad[0] else None
+        gw = x.t() @ g if ctx.needs_input_grad[1] else None
+        gb = None
+        if ctx.needs_input_grad[2]:
+            if g.is_cuda:
+                gb = (_ones_row(g.shape[0], g.device, g.dtype) @ g).reshape(-1)
+            else:
+                gb = g.sum(dim=0)
+        return gx, gw, gb
+
+
+class FastLinear(nn.Module):
+    """Drop-in linear (weight stored [in, out]) with fast bias grad."""
+
+    def __init__(self, in_dim: int, out_dim: int, bias: bool = True):
+        super().__init__()
+        self.weight = nn.Parameter(torch.empty(in_dim, out_dim))
+        self.bias = nn.Parameter(torch.zeros(out_dim)) if bias else None
+        self.reset_parameters()
+
+    def reset_parameters(self):
+        nn.init.xavier_uniform_(self.weight)
+        if self.bias is not None:
+            nn.init.zeros_(self.bias)
+
+    def forward(self, x: Tensor) -> Tensor:
+        if self.bias is None:
+            return x @ self.weight
+        return _LinearBiasFn.apply(x, self.weight, self.bias)

@@ -13,27 +13,21 @@ import torch.nn as nn
 from torch import Tensor
 
 from ..ops.dist_agg import dist_aggregate
+from .common import FastLinear
 
 
 class DistGCNConv(nn.Module):
     def __init__(self, in_dim: int, out_dim: int, layer: int, use_bias: bool = True):
         super().__init__()
         self.layer = layer
-        self.weight = nn.Parameter(torch.empty(in_dim, out_dim))
-        self.bias = nn.Parameter(torch.zeros(out_dim)) if use_bias else None
-        self.reset_parameters()
+        self.linear = FastLinear(in_dim, out_dim, bias=use_bias)
 
     def reset_parameters(self):
-        nn.init.xavier_uniform_(self.weight)
-        if self.bias is not None:
-            nn.init.zeros_(self.bias)
+        self.linear.reset_parameters()
 
     def forward(self, engine, x: Tensor) -> Tensor:
         rst = dist_aggregate(x, engine, self.layer, self.training)
-        rst = rst @ self.weight
-        if self.bias is not None:
-            rst = rst + self.bias
-        return rst
+        return self.linear(rst)
 
 
 class DistGCN(nn.Module):

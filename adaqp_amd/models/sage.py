@@ -10,6 +10,7 @@ import torch.nn as nn
 from torch import Tensor
 
 from ..ops.dist_agg import dist_aggregate
+from .common import FastLinear
 
 
 class DistSAGEConv(nn.Module):
@@ -19,19 +20,14 @@ class DistSAGEConv(nn.Module):
         assert aggregator_type in ('mean', 'gcn')
         self.layer = layer
         self.aggregator_type = aggregator_type
-        self.fc_neigh = nn.Linear(in_dim, out_dim, bias=use_bias)
+        self.fc_neigh = FastLinear(in_dim, out_dim, bias=use_bias)
         if aggregator_type == 'mean':
-            self.fc_self = nn.Linear(in_dim, out_dim, bias=use_bias)
-        self.reset_parameters()
+            self.fc_self = FastLinear(in_dim, out_dim, bias=use_bias)
 
     def reset_parameters(self):
-        nn.init.xavier_uniform_(self.fc_neigh.weight)
-        if self.fc_neigh.bias is not None:
-            nn.init.zeros_(self.fc_neigh.bias)
+        self.fc_neigh.reset_parameters()
         if self.aggregator_type == 'mean':
-            nn.init.xavier_uniform_(self.fc_self.weight)
-            if self.fc_self.bias is not None:
-                nn.init.zeros_(self.fc_self.bias)
+            self.fc_self.reset_parameters()
 
     def forward(self, engine, x: Tensor) -> Tensor:
         h_neigh = dist_aggregate(x, engine, self.layer, self.training)

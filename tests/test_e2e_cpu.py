@@ -29,15 +29,15 @@ def dense_reference(g, model_kind, agg_type, weights, num_layers, hidden):
     for i in range(num_layers):
         if model_kind == DistGNNType.DistGCN:
             agg = (A @ (h * dout.pow(-0.5)[:, None])) * din.pow(-0.5)[:, None]
-            h = agg @ weights[f'convs.{i}.weight'] + weights[f'convs.{i}.bias']
+            h = agg @ weights[f'convs.{i}.linear.weight'] + weights[f'convs.{i}.linear.bias']
         else:
             if agg_type == 'mean':
                 agg = (A @ h) / din[:, None]
-                h = h @ weights[f'convs.{i}.fc_self.weight'].t() + weights[f'convs.{i}.fc_self.bias'] \
-                    + agg @ weights[f'convs.{i}.fc_neigh.weight'].t() + weights[f'convs.{i}.fc_neigh.bias']
+                h = h @ weights[f'convs.{i}.fc_self.weight'] + weights[f'convs.{i}.fc_self.bias'] \
+                    + agg @ weights[f'convs.{i}.fc_neigh.weight'] + weights[f'convs.{i}.fc_neigh.bias']
             else:
                 agg = (A @ h + h) / (din + 1)[:, None]
-                h = agg @ weights[f'convs.{i}.fc_neigh.weight'].t() + weights[f'convs.{i}.fc_neigh.bias']
+                h = agg @ weights[f'convs.{i}.fc_neigh.weight'] + weights[f'convs.{i}.fc_neigh.bias']
         if i < num_layers - 1:
             h = torch.nn.functional.layer_norm(
                 h, (hidden,), weights[f'norms.{i}.weight'], weights[f'norms.{i}.bias'])
