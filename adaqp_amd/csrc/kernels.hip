@@ -181,7 +181,7 @@ __global__ void zero_rows_kernel(float* __restrict__ y,
         p[f] = 0.f;
 }
 
-template <int SW>
+template <int SW, bool NT>
 __global__ void spmm_csr_kernel(
     const int64_t* __restrict__ indices,
     const float* __restrict__ xl, const float* __restrict__ xr,
@@ -212,7 +212,8 @@ __global__ void spmm_csr_kernel(
             int64_t e = e0;
             if (full) {
                 for (; e + 1 < e1; e += 2) {
-                    const int64_t c0 = indices[e], c1 = indices[e + 1];
+                    const int64_t c0 = NT ? __builtin_nontemporal_load(indices + e) : indices[e];
+                    const int64_t c1 = NT ? __builtin_nontemporal_load(indices + e + 1) : indices[e + 1];
                     const float s0 = src_scale ? src_scale[c0] : 1.f;
                     const float s1 = src_scale ? src_scale[c1] : 1.f;
                     const float* p0 = (c0 < n_local ? xl + c0 * F
@@ -247,7 +248,13 @@ __global__ void spmm_csr_kernel(
             if (full && !multi) {
                 const float4 o = {(acc0.x + acc1.x) * ds, (acc0.y + acc1.y) * ds,
                                   (acc0.z + acc1.z) * ds, (acc0.w + acc1.w) * ds};
-                *reinterpret_cast<float4*>(yr) = o;
+                if (NT) {
+                    typedef float vf4 __attribute__((ext_vector_type(4)));
+                    __builtin_nontemporal_store(*reinterpret_cast<const vf4*>(&o),
+                                                reinterpret_cast<vf4*>(yr));
+                } else {
+                    *reinterpret_cast<float4*>(yr) = o;
+                }
             } else if (!multi) {
                 for (int k = 0; k < 4 && f0 + k < F; ++k)
                     yr[k] = (((float*)&acc0)[k] + ((float*)&acc1)[k]) * ds;
@@ -348,20 +355,35 @@ void spmm_csr(torch::Tensor indices, torch::Tensor xl,
     int64_t blocks = (n_seg + rows_per_block - 1) / rows_per_block;
     blocks = std::max<int64_t>(std::min<int64_t>(blocks, 16384), 1);
     const dim3 grid(blocks), block(block_threads);
-#define SPMM_LAUNCH(SW) spmm_csr_kernel<SW><<<grid, block, 0, s>>>( \
-        indices.data_ptr<int64_t>(), \
-        xl.data_ptr<float>(), xr_ptr, y.data_ptr<float>(), \
-        src_scale.numel() ? src_scale.data_ptr<float>() : nullptr, \
-        dst_scale.numel() ? dst_scale.data_ptr<float>() : nullptr, \
-        seg_row.data_ptr<int32_t>(), seg_e0.data_ptr<int64_t>(), \
-        seg_e1.data_ptr<int64_t>(), seg_multi.data_ptr<uint8_t>(), \
-        n_seg, F, n_local)
+    static const bool nt = [] {
+        const char* v = getenv("ADAQP_SPMM_NT");
+        return v && v[0] == '1';
+    }();
+    const int64_t* ind_p = indices.data_ptr<int64_t>();
+    const float* xl_p = xl.data_ptr<float>();
+    float* y_p = y.data_ptr<float>();
+    const float* ss_p = src_scale.numel() ? src_scale.data_ptr<float>() : nullptr;
+    const float* ds_p = dst_scale.numel() ? dst_scale.data_ptr<float>() : nullptr;
+    const int32_t* sr_p = seg_row.data_ptr<int32_t>();
+    const int64_t* e0_p = seg_e0.data_ptr<int64_t>();
+    const int64_t* e1_p = seg_e1.data_ptr<int64_t>();
+    const uint8_t* sm_p = seg_multi.data_ptr<uint8_t>();
+    auto dispatch = [&](auto sw_tag, auto nt_tag) {
+        constexpr int SWC = decltype(sw_tag)::value;
+        constexpr bool NTC = decltype(nt_tag)::value;
+        spmm_csr_kernel<SWC, NTC><<<grid, block, 0, s>>>(
+            ind_p, xl_p, xr_ptr, y_p, ss_p, ds_p, sr_p, e0_p, e1_p, sm_p,
+            n_seg, F, n_local);
+    };
+    using T = std::true_type; using Fa = std::false_type;
     switch (sw) {
-        case 64: SPMM_LAUNCH(64); break;
-        case 32: SPMM_LAUNCH(32); break;
-        default: SPMM_LAUNCH(16); break;
+        case 64: nt ? dispatch(std::integral_constant<int,64>{}, T{})
+                    : dispatch(std::integral_constant<int,64>{}, Fa{}); break;
+        case 32: nt ? dispatch(std::integral_constant<int,32>{}, T{})
+                    : dispatch(std::integral_constant<int,32>{}, Fa{}); break;
+        default: nt ? dispatch(std::integral_constant<int,16>{}, T{})
+                    : dispatch(std::integral_constant<int,16>{}, Fa{}); break;
     }
-#undef SPMM_LAUNCH
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

@@ -90,7 +90,14 @@ class Trainer:
                                 scale=scale)
                 save_partitions(partition_all(g, world), part_dir, ds)
             self.comm.barrier()
-        self.graph = load_partition(part_dir, ds, world, rank)
+        try:
+            self.graph = load_partition(part_dir, ds, world, rank)
+        except FileNotFoundError as e:
+            raise SystemExit(
+                f'partition for {ds} with {world} parts not found under '
+                f'{part_dir} (run graph_partition.py --dataset {ds} '
+                f'--partition_size {world}, or launch with a matching '
+                f'--nproc-per-node): {e}')
 
     def _set_engine(self):
         m = self.cfg['model']
