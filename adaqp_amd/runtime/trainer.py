@@ -69,9 +69,13 @@ class Trainer:
 
     # ------------------------------------------------------------------
     def _setup_logging(self, level: str):
-        logging.basicConfig(
-            level=getattr(logging, level.upper(), logging.INFO),
-            format=f'[rank {self.comm.rank}] %(asctime)s %(levelname)s %(message)s')
+        fmt = f'[rank {self.comm.rank}] %(asctime)s %(levelname)s %(message)s'
+        logging.basicConfig(level=getattr(logging, level.upper(), logging.INFO),
+                            format=fmt)
+        # file log, reference parity (runtime_util.py:22-32)
+        fh = logging.FileHandler('trainer.log')
+        fh.setFormatter(logging.Formatter(fmt))
+        logging.getLogger('trainer').addHandler(fh)
 
     def _set_graph(self):
         args = self.args

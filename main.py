@@ -29,6 +29,8 @@ def main():
                    help='torch.distributed backend (default: RCCL on GPU, gloo on CPU)')
     p.add_argument('--init_method', type=str, default='env://')
     p.add_argument('--logger_level', type=str, default='INFO')
+    p.add_argument('--num_parts', type=int, default=None,
+                   help='expected partition count; validated against WORLD_SIZE')
     p.add_argument('--partition_dir', type=str, default='part_data')
     p.add_argument('--num_epochs', type=int, default=None)
     p.add_argument('--lr', type=float, default=None)
@@ -46,6 +48,11 @@ def main():
     os.environ.setdefault('RANK', '0')
     os.environ.setdefault('WORLD_SIZE', '1')
     os.environ.setdefault('LOCAL_RANK', os.environ.get('RANK', '0'))
+
+    ws = int(os.environ.get('WORLD_SIZE', '1'))
+    if args.num_parts is not None and args.num_parts != ws:
+        raise SystemExit(f'--num_parts {args.num_parts} != WORLD_SIZE {ws}: '
+                         'launch with --nproc-per-node equal to num_parts')
 
     trainer = Trainer(args)
     best = trainer.train()
