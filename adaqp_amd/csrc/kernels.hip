@@ -19,6 +19,7 @@
 //
 // Built for gfx950 only. No CUDA compatibility path.
 #include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
@@ -53,6 +54,15 @@ __device__ __forceinline__ float bf16_to_f32(uint16_t h) {
     return __uint_as_float(((uint32_t)h) << 16);
 }
 
+// T = float (V=4 feats/lane) or ushort bf16 bits (V=8 feats/lane); fp32 accumulate.
+template <typename T>
+__device__ __forceinline__ float to_f32(T v);
+template <> __device__ __forceinline__ float to_f32<float>(float v) { return v; }
+template <> __device__ __forceinline__ float to_f32<ushort>(ushort v) {
+    return bf16_to_f32(v);
+}
+
+
 // ---------------------------------------------------------------------------
 // quant_pack: one wave per node.
 //   rows[i]   : row of x to quantize
@@ -61,9 +71,9 @@ __device__ __forceinline__ float bf16_to_f32(uint16_t h) {
 // Lane l owns features [l*vpb + k*WAVE*vpb, ...): it loads vpb consecutive
 // floats, reduces min/max across the wave, then emits one byte per k.
 // ---------------------------------------------------------------------------
-template <int BITS>
+template <int BITS, typename T>
 __global__ void quant_pack_kernel(
-    const float* __restrict__ x, const int64_t* __restrict__ rows,
+    const T* __restrict__ x, const int64_t* __restrict__ rows,
     const int64_t* __restrict__ pos, const int64_t* __restrict__ off,
     int64_t n, int64_t F, int64_t ld, uint32_t seed,
     uint8_t* __restrict__ payload, uint16_t* __restrict__ params) {
@@ -73,7 +83,7 @@ __global__ void quant_pack_kernel(
     if (wid >= n) return;
     const int lane = threadIdx.x & (WAVE - 1);
     const int64_t row = rows[wid];
-    const float* xr = x + row * ld;
+    const T* xr = x + row * ld;
     const int64_t bpn = (F * BITS + 7) / 8;
 
     // pass 1: min/max (coalesced vpb-wide vector loads)
@@ -82,7 +92,7 @@ __global__ void quant_pack_kernel(
 #pragma unroll
         for (int k = 0; k < VPB; ++k) {
             if (f0 + k < F) {
-                float v = xr[f0 + k];
+                float v = to_f32<T>(xr[f0 + k]);
                 mn = fminf(mn, v);
                 mx = fmaxf(mx, v);
             }
@@ -114,7 +124,7 @@ __global__ void quant_pack_kernel(
         for (int k = 0; k < VPB; ++k) {
             const int64_t f = b0 * VPB + k;
             if (f < F && scale > 0.f) {
-                float v = (xr[f] - rmin) * scale;
+                float v = (to_f32<T>(xr[f]) - rmin) * scale;
                 float u = uniform01(seed, tag, (uint32_t)f);
                 int q = (int)floorf(v + u);
                 q = max(0, min(q, (1 << BITS) - 1));
@@ -128,12 +138,13 @@ __global__ void quant_pack_kernel(
 // ---------------------------------------------------------------------------
 // quant_unpack: one wave per node, fused scatter into out[rows[i]].
 // ---------------------------------------------------------------------------
-template <int BITS>
+template <int BITS, typename T>
 __global__ void quant_unpack_kernel(
     const uint8_t* __restrict__ payload, const uint16_t* __restrict__ params,
     const int64_t* __restrict__ rows, const int64_t* __restrict__ pos,
     const int64_t* __restrict__ off, int64_t n, int64_t F, int64_t ld,
-    float* __restrict__ out) {
+    T* __restrict__ out) {
+    constexpr bool BF = sizeof(T) == 2;
     constexpr int VPB = 8 / BITS;
     const int64_t wid = (int64_t)blockIdx.x * (blockDim.x / WAVE)
                       + threadIdx.x / WAVE;
@@ -144,7 +155,7 @@ __global__ void quant_unpack_kernel(
     const float rmin = bf16_to_f32(params[2 * p + 1]);
     const float inv = scale > 0.f ? 1.f / scale : 0.f;
     const uint8_t* in = payload + off[wid];
-    float* o = out + rows[wid] * ld;
+    T* o = out + rows[wid] * ld;
     for (int64_t b0 = lane; b0 * VPB < F; b0 += WAVE) {
         const uint32_t byte = in[b0];
 #pragma unroll
@@ -152,7 +163,9 @@ __global__ void quant_unpack_kernel(
             const int64_t f = b0 * VPB + k;
             if (f < F) {
                 const int q = (byte >> (k * BITS)) & ((1 << BITS) - 1);
-                o[f] = (float)q * inv + rmin;
+                const float v = (float)q * inv + rmin;
+                if constexpr (BF) o[f] = f32_to_bf16(v);
+                else o[f] = v;
             }
         }
     }
@@ -170,26 +183,30 @@ __global__ void quant_unpack_kernel(
 // XCD-aware bijective block swizzle gives each XCD a contiguous row chunk
 // so clustered neighbor rows hit the same L2 (guide §5.5 T1).
 // ---------------------------------------------------------------------------
-__global__ void zero_rows_kernel(float* __restrict__ y,
+template <typename T>
+__global__ void zero_rows_kernel(T* __restrict__ y,
                                  const int32_t* __restrict__ rows,
                                  int64_t n, int64_t F) {
     const int64_t wid = (int64_t)blockIdx.x * (blockDim.x / WAVE)
                       + threadIdx.x / WAVE;
     if (wid >= n) return;
-    float* p = y + (int64_t)rows[wid] * F;
+    T* p = y + (int64_t)rows[wid] * F;
     for (int64_t f = threadIdx.x & (WAVE - 1); f < F; f += WAVE)
-        p[f] = 0.f;
+        p[f] = T(0);
 }
 
-template <int SW, bool NT>
+template <int SW, bool NT, typename T>
 __global__ void spmm_csr_kernel(
     const int64_t* __restrict__ indices,
-    const float* __restrict__ xl, const float* __restrict__ xr,
-    float* __restrict__ y,
+    const T* __restrict__ xl, const T* __restrict__ xr,
+    T* __restrict__ y,
     const float* __restrict__ src_scale, const float* __restrict__ dst_scale,
     const int32_t* __restrict__ seg_row, const int64_t* __restrict__ seg_e0,
     const int64_t* __restrict__ seg_e1, const uint8_t* __restrict__ seg_multi,
     int64_t n_seg, int64_t F, int64_t n_local) {
+    constexpr bool BF = sizeof(T) == 2;
+    constexpr int V = 16 / sizeof(T);          // features per lane (16B loads)
+    typedef float vf4 __attribute__((ext_vector_type(4)));
     const int rows_per_block = blockDim.x / SW;
     // bijective XCD swizzle: blocks [0,nwg) -> xcd-contiguous chunks
     const int64_t nwg = gridDim.x;
@@ -205,10 +222,11 @@ __global__ void spmm_csr_kernel(
         const int64_t e0 = seg_e0[it], e1 = seg_e1[it];
         const bool multi = seg_multi[it];
         const float ds = dst_scale ? dst_scale[r] : 1.f;
-        for (int64_t f0 = (int64_t)sl * 4; f0 < F; f0 += (int64_t)SW * 4) {
-            float4 acc0 = {0.f, 0.f, 0.f, 0.f};
-            float4 acc1 = {0.f, 0.f, 0.f, 0.f};
-            const bool full = (f0 + 4 <= F);
+        for (int64_t f0 = (int64_t)sl * V; f0 < F; f0 += (int64_t)SW * V) {
+            float acc0[V], acc1[V];
+#pragma unroll
+            for (int k = 0; k < V; ++k) { acc0[k] = 0.f; acc1[k] = 0.f; }
+            const bool full = (f0 + V <= F);
             int64_t e = e0;
             if (full) {
                 for (; e + 1 < e1; e += 2) {
@@ -216,52 +234,65 @@ __global__ void spmm_csr_kernel(
                     const int64_t c1 = NT ? __builtin_nontemporal_load(indices + e + 1) : indices[e + 1];
                     const float s0 = src_scale ? src_scale[c0] : 1.f;
                     const float s1 = src_scale ? src_scale[c1] : 1.f;
-                    const float* p0 = (c0 < n_local ? xl + c0 * F
-                                                    : xr + (c0 - n_local) * F) + f0;
-                    const float* p1 = (c1 < n_local ? xl + c1 * F
-                                                    : xr + (c1 - n_local) * F) + f0;
-                    const float4 v0 = *reinterpret_cast<const float4*>(p0);
-                    const float4 v1 = *reinterpret_cast<const float4*>(p1);
-                    acc0.x = fmaf(v0.x, s0, acc0.x); acc1.x = fmaf(v1.x, s1, acc1.x);
-                    acc0.y = fmaf(v0.y, s0, acc0.y); acc1.y = fmaf(v1.y, s1, acc1.y);
-                    acc0.z = fmaf(v0.z, s0, acc0.z); acc1.z = fmaf(v1.z, s1, acc1.z);
-                    acc0.w = fmaf(v0.w, s0, acc0.w); acc1.w = fmaf(v1.w, s1, acc1.w);
+                    const T* p0 = (c0 < n_local ? xl + c0 * F : xr + (c0 - n_local) * F) + f0;
+                    const T* p1 = (c1 < n_local ? xl + c1 * F : xr + (c1 - n_local) * F) + f0;
+                    const uint4 r0 = *reinterpret_cast<const uint4*>(p0);
+                    const uint4 r1 = *reinterpret_cast<const uint4*>(p1);
+                    const T* v0 = reinterpret_cast<const T*>(&r0);
+                    const T* v1 = reinterpret_cast<const T*>(&r1);
+#pragma unroll
+                    for (int k = 0; k < V; ++k) {
+                        acc0[k] = fmaf(to_f32<T>(v0[k]), s0, acc0[k]);
+                        acc1[k] = fmaf(to_f32<T>(v1[k]), s1, acc1[k]);
+                    }
                 }
             }
             for (; e < e1; ++e) {
                 const int64_t c = indices[e];
                 const float s = src_scale ? src_scale[c] : 1.f;
-                const float* xc = (c < n_local ? xl + c * F
-                                               : xr + (c - n_local) * F) + f0;
-                if (full) {
-                    const float4 v = *reinterpret_cast<const float4*>(xc);
-                    acc0.x = fmaf(v.x, s, acc0.x);
-                    acc0.y = fmaf(v.y, s, acc0.y);
-                    acc0.z = fmaf(v.z, s, acc0.z);
-                    acc0.w = fmaf(v.w, s, acc0.w);
-                } else {
-                    for (int k = 0; k < 4 && f0 + k < F; ++k)
-                        ((float*)&acc0)[k] = fmaf(xc[k], s, ((float*)&acc0)[k]);
-                }
+                const T* xc = (c < n_local ? xl + c * F : xr + (c - n_local) * F) + f0;
+#pragma unroll
+                for (int k = 0; k < V; ++k)
+                    if (f0 + k < F)
+                        acc0[k] = fmaf(to_f32<T>(xc[k]), s, acc0[k]);
             }
-            float* yr = y + r * F + f0;
-            if (full && !multi) {
-                const float4 o = {(acc0.x + acc1.x) * ds, (acc0.y + acc1.y) * ds,
-                                  (acc0.z + acc1.z) * ds, (acc0.w + acc1.w) * ds};
-                if (NT) {
-                    typedef float vf4 __attribute__((ext_vector_type(4)));
-                    __builtin_nontemporal_store(*reinterpret_cast<const vf4*>(&o),
-                                                reinterpret_cast<vf4*>(yr));
-                } else {
-                    *reinterpret_cast<float4*>(yr) = o;
+            T* yr = y + r * F + f0;
+            if (!multi) {
+                T outv[V];
+#pragma unroll
+                for (int k = 0; k < V; ++k) {
+                    const float o = (acc0[k] + acc1[k]) * ds;
+                    if constexpr (BF) outv[k] = f32_to_bf16(o);
+                    else outv[k] = o;
                 }
-            } else if (!multi) {
-                for (int k = 0; k < 4 && f0 + k < F; ++k)
-                    yr[k] = (((float*)&acc0)[k] + ((float*)&acc1)[k]) * ds;
+                if (full) {
+                    if (NT)
+                        __builtin_nontemporal_store(*reinterpret_cast<const vf4*>(outv),
+                                                    reinterpret_cast<vf4*>(yr));
+                    else
+                        *reinterpret_cast<uint4*>(yr) = *reinterpret_cast<const uint4*>(outv);
+                } else {
+                    for (int k = 0; k < V && f0 + k < F; ++k)
+                        yr[k] = outv[k];
+                }
             } else {
                 // long row split across segments: combine via atomics
-                for (int k = 0; k < 4 && f0 + k < F; ++k)
-                    atomicAdd(yr + k, (((float*)&acc0)[k] + ((float*)&acc1)[k]) * ds);
+                if constexpr (BF) {
+#pragma unroll
+                    for (int k = 0; k < V; k += 2) {
+                        if (f0 + k < F) {
+                            __hip_bfloat162 v;
+                            v.x = __float2bfloat16((acc0[k] + acc1[k]) * ds);
+                            v.y = __float2bfloat16(f0 + k + 1 < F
+                                                   ? (acc0[k + 1] + acc1[k + 1]) * ds : 0.f);
+                            unsafeAtomicAdd(reinterpret_cast<__hip_bfloat162*>(yr + k), v);
+                        }
+                    }
+                } else {
+                    for (int k = 0; k < V && f0 + k < F; ++k)
+                        atomicAdd(reinterpret_cast<float*>(yr) + k,
+                                  (acc0[k] + acc1[k]) * ds);
+                }
             }
         }
     }
@@ -278,7 +309,9 @@ void quant_pack(torch::Tensor x, torch::Tensor rows, torch::Tensor pos,
                 torch::Tensor off, int64_t bits, int64_t seed,
                 torch::Tensor payload, torch::Tensor params) {
     CHECK_DEV(x); CHECK_CONTIG(x); CHECK_DEV(rows); CHECK_DEV(payload); CHECK_DEV(params);
-    TORCH_CHECK(x.scalar_type() == torch::kFloat32, "quant_pack expects fp32");
+    const bool bf16 = x.scalar_type() == torch::kBFloat16;
+    TORCH_CHECK(bf16 || x.scalar_type() == torch::kFloat32,
+                "quant_pack expects fp32 or bf16");
     const int64_t n = rows.numel();
     if (n == 0) return;
     const int64_t F = x.size(1), ld = x.stride(0);
@@ -286,14 +319,15 @@ void quant_pack(torch::Tensor x, torch::Tensor rows, torch::Tensor pos,
     const dim3 block(WAVE * waves_per_block);
     const dim3 grid((n + waves_per_block - 1) / waves_per_block);
     auto s = cur_stream();
-#define LAUNCH(B) quant_pack_kernel<B><<<grid, block, 0, s>>>( \
-        x.data_ptr<float>(), rows.data_ptr<int64_t>(), pos.data_ptr<int64_t>(), \
+#define LAUNCH(B, T) quant_pack_kernel<B, T><<<grid, block, 0, s>>>( \
+        reinterpret_cast<const T*>(x.data_ptr()), rows.data_ptr<int64_t>(), \
+        pos.data_ptr<int64_t>(), \
         off.data_ptr<int64_t>(), n, F, ld, (uint32_t)seed, \
         payload.data_ptr<uint8_t>(), reinterpret_cast<uint16_t*>(params.data_ptr<at::BFloat16>()))
     switch (bits) {
-        case 2: LAUNCH(2); break;
-        case 4: LAUNCH(4); break;
-        case 8: LAUNCH(8); break;
+        case 2: if (bf16) LAUNCH(2, ushort); else LAUNCH(2, float); break;
+        case 4: if (bf16) LAUNCH(4, ushort); else LAUNCH(4, float); break;
+        case 8: if (bf16) LAUNCH(8, ushort); else LAUNCH(8, float); break;
         default: TORCH_CHECK(false, "bits must be 2/4/8");
     }
 #undef LAUNCH
@@ -303,22 +337,24 @@ void quant_unpack(torch::Tensor payload, torch::Tensor params, torch::Tensor row
                   torch::Tensor pos, torch::Tensor off, int64_t bits, int64_t F,
                   torch::Tensor out) {
     CHECK_DEV(payload); CHECK_DEV(out); CHECK_CONTIG(out);
-    TORCH_CHECK(out.scalar_type() == torch::kFloat32, "quant_unpack expects fp32 out");
+    const bool bf16 = out.scalar_type() == torch::kBFloat16;
+    TORCH_CHECK(bf16 || out.scalar_type() == torch::kFloat32,
+                "quant_unpack expects fp32 or bf16 out");
     const int64_t n = rows.numel();
     if (n == 0) return;
     const int waves_per_block = 4;
     const dim3 block(WAVE * waves_per_block);
     const dim3 grid((n + waves_per_block - 1) / waves_per_block);
     auto s = cur_stream();
-#define LAUNCH(B) quant_unpack_kernel<B><<<grid, block, 0, s>>>( \
+#define LAUNCH(B, T) quant_unpack_kernel<B, T><<<grid, block, 0, s>>>( \
         payload.data_ptr<uint8_t>(), \
         reinterpret_cast<uint16_t*>(params.data_ptr<at::BFloat16>()), \
         rows.data_ptr<int64_t>(), pos.data_ptr<int64_t>(), off.data_ptr<int64_t>(), \
-        n, F, out.stride(0), out.data_ptr<float>())
+        n, F, out.stride(0), reinterpret_cast<T*>(out.data_ptr()))
     switch (bits) {
-        case 2: LAUNCH(2); break;
-        case 4: LAUNCH(4); break;
-        case 8: LAUNCH(8); break;
+        case 2: if (bf16) LAUNCH(2, ushort); else LAUNCH(2, float); break;
+        case 4: if (bf16) LAUNCH(4, ushort); else LAUNCH(4, float); break;
+        case 8: if (bf16) LAUNCH(8, ushort); else LAUNCH(8, float); break;
         default: TORCH_CHECK(false, "bits must be 2/4/8");
     }
 #undef LAUNCH
@@ -330,26 +366,37 @@ void spmm_csr(torch::Tensor indices, torch::Tensor xl,
               torch::Tensor seg_e0, torch::Tensor seg_e1,
               torch::Tensor seg_multi, torch::Tensor zero_rows) {
     CHECK_DEV(xl); CHECK_CONTIG(xl); CHECK_DEV(y); CHECK_CONTIG(y);
-    TORCH_CHECK(xl.scalar_type() == torch::kFloat32, "spmm_csr expects fp32");
+    const bool bf16 = xl.scalar_type() == torch::kBFloat16;
+    TORCH_CHECK(bf16 || xl.scalar_type() == torch::kFloat32,
+                "spmm_csr expects fp32 or bf16");
+    TORCH_CHECK(y.scalar_type() == xl.scalar_type(), "x/y dtype mismatch");
     const int64_t F = xl.size(1);
     const int64_t n_local = xl.size(0);
     const int64_t n_seg = seg_row.numel();
-    const float* xr_ptr = nullptr;
+    const void* xr_ptr = nullptr;
     if (xr.numel()) {
         CHECK_DEV(xr); CHECK_CONTIG(xr);
         TORCH_CHECK(xr.size(1) == F, "remote feature dim mismatch");
-        xr_ptr = xr.data_ptr<float>();
+        TORCH_CHECK(xr.scalar_type() == xl.scalar_type(), "x dtype mismatch");
+        xr_ptr = xr.data_ptr();
     }
     auto s = cur_stream();
     if (zero_rows.numel()) {
         const dim3 zb(256), zg((zero_rows.numel() + 3) / 4);
-        zero_rows_kernel<<<zg, zb, 0, s>>>(y.data_ptr<float>(),
-                                           zero_rows.data_ptr<int32_t>(),
-                                           zero_rows.numel(), F);
+        if (bf16)
+            zero_rows_kernel<ushort><<<zg, zb, 0, s>>>(
+                reinterpret_cast<ushort*>(y.data_ptr()),
+                zero_rows.data_ptr<int32_t>(), zero_rows.numel(), F);
+        else
+            zero_rows_kernel<float><<<zg, zb, 0, s>>>(
+                y.data_ptr<float>(), zero_rows.data_ptr<int32_t>(),
+                zero_rows.numel(), F);
     }
     if (n_seg == 0) return;
-    // sub-wavefront width from F: keep all 64 lanes of a wave busy
-    const int sw = F > 128 ? 64 : (F > 64 ? 32 : 16);
+    // sub-wavefront width: smallest of {16,32,64} covering F with 16B lanes
+    const int vpl = bf16 ? 8 : 4;            // features per lane
+    int sw = 16;
+    while (sw < 64 && sw * vpl < F) sw *= 2;
     const int block_threads = WAVE * 4;
     const int rows_per_block = block_threads / sw;
     int64_t blocks = (n_seg + rows_per_block - 1) / rows_per_block;
@@ -360,29 +407,37 @@ void spmm_csr(torch::Tensor indices, torch::Tensor xl,
         return v && v[0] == '1';
     }();
     const int64_t* ind_p = indices.data_ptr<int64_t>();
-    const float* xl_p = xl.data_ptr<float>();
-    float* y_p = y.data_ptr<float>();
     const float* ss_p = src_scale.numel() ? src_scale.data_ptr<float>() : nullptr;
     const float* ds_p = dst_scale.numel() ? dst_scale.data_ptr<float>() : nullptr;
     const int32_t* sr_p = seg_row.data_ptr<int32_t>();
     const int64_t* e0_p = seg_e0.data_ptr<int64_t>();
     const int64_t* e1_p = seg_e1.data_ptr<int64_t>();
     const uint8_t* sm_p = seg_multi.data_ptr<uint8_t>();
-    auto dispatch = [&](auto sw_tag, auto nt_tag) {
+    auto dispatch = [&](auto sw_tag, auto nt_tag, auto t_tag) {
         constexpr int SWC = decltype(sw_tag)::value;
         constexpr bool NTC = decltype(nt_tag)::value;
-        spmm_csr_kernel<SWC, NTC><<<grid, block, 0, s>>>(
-            ind_p, xl_p, xr_ptr, y_p, ss_p, ds_p, sr_p, e0_p, e1_p, sm_p,
-            n_seg, F, n_local);
+        using TC = typename decltype(t_tag)::type;
+        spmm_csr_kernel<SWC, NTC, TC><<<grid, block, 0, s>>>(
+            ind_p, reinterpret_cast<const TC*>(xl.data_ptr()),
+            reinterpret_cast<const TC*>(xr_ptr),
+            reinterpret_cast<TC*>(y.data_ptr()), ss_p, ds_p,
+            sr_p, e0_p, e1_p, sm_p, n_seg, F, n_local);
     };
-    using T = std::true_type; using Fa = std::false_type;
+    auto dis2 = [&](auto sw_tag) {
+        struct FT { using type = float; };
+        struct BT { using type = ushort; };
+        if (bf16) {
+            if (nt) dispatch(sw_tag, std::true_type{}, BT{});
+            else dispatch(sw_tag, std::false_type{}, BT{});
+        } else {
+            if (nt) dispatch(sw_tag, std::true_type{}, FT{});
+            else dispatch(sw_tag, std::false_type{}, FT{});
+        }
+    };
     switch (sw) {
-        case 64: nt ? dispatch(std::integral_constant<int,64>{}, T{})
-                    : dispatch(std::integral_constant<int,64>{}, Fa{}); break;
-        case 32: nt ? dispatch(std::integral_constant<int,32>{}, T{})
-                    : dispatch(std::integral_constant<int,32>{}, Fa{}); break;
-        default: nt ? dispatch(std::integral_constant<int,16>{}, T{})
-                    : dispatch(std::integral_constant<int,16>{}, Fa{}); break;
+        case 64: dis2(std::integral_constant<int,64>{}); break;
+        case 32: dis2(std::integral_constant<int,32>{}); break;
+        default: dis2(std::integral_constant<int,16>{}); break;
     }
 }
 

@@ -62,6 +62,14 @@ class FastLinear(nn.Module):
             nn.init.zeros_(self.bias)
 
     def forward(self, x: Tensor) -> Tensor:
-        if self.bias is None:
-            return x @ self.weight
-        return _LinearBiasFn.apply(x, self.weight, self.bias)
+        w, b = self.weight, self.bias
+        if torch.is_autocast_enabled('cuda') and x.is_cuda:
+            # differentiable casts OUTSIDE the Function: bf16 compute,
+            # fp32 master weights get fp32 grads via the cast backward
+            dt = torch.get_autocast_dtype('cuda')
+            x = x.to(dt)
+            w = w.to(dt)
+            b = b.to(dt) if b is not None else None
+        if b is None:
+            return x @ w
+        return _LinearBiasFn.apply(x, w, b)

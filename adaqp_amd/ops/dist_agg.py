@@ -106,9 +106,9 @@ def _agg(engine, view, x_local: Tensor, x_remote, src_scale, dst_scale) -> Tenso
 def _self_term(engine, x_local: Tensor, mode: PropagationMode) -> Tensor:
     """SAGE 'gcn' aggregator self term (reference ops.py:44-47,57-64)."""
     I = engine.graph.num_inner
-    if mode == PropagationMode.Forward:
-        return x_local[:I] * engine.sage1_dst_f[:, None]
-    return x_local[:I] * engine.sage1_src_b[:I, None]
+    scale = (engine.sage1_dst_f if mode == PropagationMode.Forward
+             else engine.sage1_src_b[:I])
+    return (x_local[:I] * scale[:, None]).to(x_local.dtype)
 
 
 # --------------------------------------------------------------------------
@@ -184,13 +184,15 @@ class DistAgg(torch.autograd.Function):
     def forward(ctx, x_local: Tensor, engine, layer: int, is_train: bool) -> Tensor:
         ctx.engine = engine
         ctx.layer = layer
+        ctx.in_dtype = x_local.dtype
+        x_local = x_local.to(engine.compute_dtype).contiguous()
         return propagate(engine, x_local, layer, is_train, PropagationMode.Forward)
 
     @staticmethod
     def backward(ctx, grad_out: Tensor):
-        gx = propagate(ctx.engine, grad_out.contiguous(), ctx.layer, True,
-                       PropagationMode.Backward)
-        return gx, None, None, None
+        g = grad_out.to(ctx.engine.compute_dtype).contiguous()
+        gx = propagate(ctx.engine, g, ctx.layer, True, PropagationMode.Backward)
+        return gx.to(ctx.in_dtype), None, None, None
 
 
 def dist_aggregate(x_local: Tensor, engine, layer: int, is_train: bool) -> Tensor:

@@ -47,6 +47,7 @@ class GraphEngine:
         self.msg_dims = list(msg_dims)          # feature dim per layer exchange
         self.num_layers = len(msg_dims)
         self.base_seed = base_seed
+        self.compute_dtype = torch.float32   # set to torch.bfloat16 for bf16 mode
         self._rng_counter = 0
         self.timer = Timer(enabled=False, cuda=self.device.type == 'cuda')
         self.is_tracing = False

@@ -114,6 +114,8 @@ class Trainer:
                                   device=self.comm.device)
         if getattr(self.args, 'time_breakdown', False):
             self.engine.timer.enabled = True
+        if getattr(self.args, 'dtype', 'fp32') == 'bf16':
+            self.engine.compute_dtype = torch.bfloat16
 
     def _set_assigner(self):
         a = self.cfg['assignment']

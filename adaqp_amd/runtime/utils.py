@@ -33,12 +33,21 @@ def compute_loss(logits: Tensor, labels: Tensor, mask: Tensor,
     return raw / global_count
 
 
+def _autocast(engine):
+    use = (engine.compute_dtype == torch.bfloat16
+           and engine.device.type == 'cuda')
+    return torch.autocast('cuda', dtype=torch.bfloat16, enabled=use) if use \
+        else torch.autocast('cpu', enabled=False)
+
+
 def train_epoch(engine, model, optimizer, global_count: Tensor,
                 multilabel: bool) -> Tensor:
     comm = Communicator.ctx
     model.train()
     optimizer.zero_grad(set_to_none=False)
-    logits = model(engine, engine.graph.feats)
+    with _autocast(engine):
+        logits = model(engine, engine.graph.feats)
+    logits = logits.float()
     loss = compute_loss(logits, engine.graph.labels, engine.graph.train_mask,
                         multilabel, global_count)
     loss.backward()
@@ -54,7 +63,9 @@ def evaluate(engine, model, multilabel: bool) -> Dict[str, float]:
     micro-F1 (multilabel), all-reduced counts (``runtime_util.py:139-197``)."""
     comm = Communicator.ctx
     model.eval()
-    logits = model(engine, engine.graph.feats)
+    with _autocast(engine):
+        logits = model(engine, engine.graph.feats)
+    logits = logits.float()
     g = engine.graph
     out = {}
     if multilabel:

@@ -38,6 +38,9 @@ def parse_args():
                    help='uniform bit width for quantized modes')
     p.add_argument('--part-dir', type=str, default='part_data_bench')
     p.add_argument('--cpu', action='store_true', help='force CPU (debug)')
+    p.add_argument('--dtype', type=str, default='fp32', choices=['fp32', 'bf16'],
+                   help='compute dtype (fp32 matches the reference; bf16 '
+                        'halves activation traffic, fp32 master weights)')
     return p.parse_args()
 
 
@@ -94,6 +97,8 @@ def main():
     msg_dims = [feat_dim] + [args.hidden] * (args.layers - 1)
     engine = GraphEngine(lg, mode, mtype, msg_dims, agg_type='mean',
                          device=comm.device)
+    if args.dtype == 'bf16':
+        engine.compute_dtype = torch.bfloat16
     if mode.bit_type.name == 'QUANT':
         engine.set_uniform_assignment(args.assign_bits)
 
@@ -142,7 +147,7 @@ def main():
             'higher_is_better': False,
             'scaling': 'strong',
             'vs_baseline': None,
-            'dtype': 'fp32',
+            'dtype': args.dtype,
             'data': f'synthetic ({args.dataset} shape: {int(n*args.scale)} nodes, '
                     f'~{int(e*args.scale)} edges, planted METIS-like locality), random-init weights',
             'config': {

@@ -36,6 +36,7 @@ def main():
     p.add_argument('--lr', type=float, default=None)
     p.add_argument('--log_steps', type=int, default=None)
     p.add_argument('--seed', type=int, default=None)
+    p.add_argument('--dtype', type=str, default='fp32', choices=['fp32', 'bf16'])
     p.add_argument('--time_breakdown', action='store_true',
                    help='enable per-epoch comm/quant/agg span timing '
                         '(adds sync fences; use rocprofv3 for kernel evidence)')

@@ -184,3 +184,76 @@ def test_e2e_train_step_gpu():
         assert l < l0  # training must reduce loss on a learnable graph
     finally:
         Communicator.shutdown()
+
+
+def test_spmm_bf16_matches_fp32(etol=0.02):
+    from adaqp_amd.graph import random_partitioned_graph, partition_all
+    from adaqp_amd.ops.kernels import SpmmView, spmm
+    _native()
+    torch.manual_seed(6)
+    g = random_partitioned_graph(800, 8000, 4, 3, 1, seed=12)
+    lg = partition_all(g, 1)[0]
+    for F in (100, 256):
+        x = torch.randn(lg.num_nodes, F, device='cuda')
+        src = torch.rand(lg.num_nodes, device='cuda') + 0.5
+        dst = torch.rand(lg.num_inner, device='cuda') + 0.5
+        view = SpmmView(lg.indptr.cuda(), lg.indices.cuda(), 0, lg.num_inner).to('cuda')
+        y32 = spmm(view, x, None, src, dst)
+        y16 = spmm(view, x.bfloat16(), None, src, dst)
+        assert y16.dtype == torch.bfloat16
+        scale = y32.abs().max()
+        err = (y16.float() - y32).abs().max() / scale
+        assert err < etol, (F, float(err))
+
+
+def test_quant_roundtrip_bf16():
+    from adaqp_amd.comm.buffers import bytes_per_node
+    C = _native()
+    torch.manual_seed(7)
+    n, F, bits = 32, 256, 8
+    x = torch.randn(n, F, device='cuda', dtype=torch.bfloat16)
+    rows = torch.arange(n, dtype=torch.int64, device='cuda')
+    bpn = bytes_per_node(F, bits)
+    off = rows * bpn
+    payload = torch.zeros(n * bpn, dtype=torch.uint8, device='cuda')
+    params = torch.zeros(2 * n, dtype=torch.bfloat16, device='cuda')
+    C.quant_pack(x, rows, rows, off, bits, 11, payload, params)
+    out = torch.zeros(n, F, device='cuda', dtype=torch.bfloat16)
+    C.quant_unpack(payload, params, rows, rows, off, bits, F, out)
+    xf = x.float()
+    rng = xf.max(1).values - xf.min(1).values
+    step = rng / (2 ** bits - 1)
+    err = (out.float() - xf).abs().max(1).values
+    assert (err <= step * 1.05 + rng * 0.02).all()
+
+
+def test_e2e_train_step_bf16():
+    import os
+    os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
+    os.environ.setdefault('MASTER_PORT', '29785')
+    os.environ.setdefault('RANK', '0')
+    os.environ.setdefault('WORLD_SIZE', '1')
+    from adaqp_amd.comm import Communicator
+    from adaqp_amd.runtime import GraphEngine
+    from adaqp_amd.runtime.utils import train_epoch, global_train_count
+    from adaqp_amd.models import DistSAGE
+    from adaqp_amd.helpers import RunMode, DistGNNType
+    from adaqp_amd.graph import random_partitioned_graph, partition_all
+    if Communicator.ctx is None:
+        comm = Communicator()
+    else:
+        comm = Communicator.ctx
+    g = random_partitioned_graph(3000, 30000, 32, 7, 1, seed=5,
+                                 teacher_labels=True)
+    lg = partition_all(g, 1)[0]
+    engine = GraphEngine(lg, RunMode('AdaQP'), DistGNNType.DistSAGE,
+                         msg_dims=[32, 16, 16], device=comm.device)
+    engine.compute_dtype = torch.bfloat16
+    engine.set_uniform_assignment(8)
+    model = DistSAGE(32, 16, 7, 3, dropout=0.0).to(comm.device)
+    opt = torch.optim.Adam(model.parameters(), lr=0.01)
+    gc = global_train_count(engine)
+    l0 = float(train_epoch(engine, model, opt, gc, False))
+    for _ in range(25):
+        l = float(train_epoch(engine, model, opt, gc, False))
+    assert torch.isfinite(torch.tensor(l)) and l < l0
