@@ -5,9 +5,6 @@ DistAggConv then applies the linear transform; stack uses
 dropout -> LayerNorm -> ReLU between layers; Xavier init)."""
 from __future__ import annotations
 
-import math
-from typing import List
-
 import torch
 import torch.nn as nn
 from torch import Tensor

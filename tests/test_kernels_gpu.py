@@ -257,3 +257,4 @@ def test_e2e_train_step_bf16():
     for _ in range(25):
         l = float(train_epoch(engine, model, opt, gc, False))
     assert torch.isfinite(torch.tensor(l)) and l < l0
+    Communicator.shutdown()
