@@ -195,7 +195,16 @@ __global__ void zero_rows_kernel(T* __restrict__ y,
         p[f] = T(0);
 }
 
-template <int SW, bool NT, typename T>
+// VE = elements per lane per load; chosen host-side as the largest of
+// {8,4,2}(bf16) / {4,2,1}(f32) dividing F, so every row-base load of
+// width VE*sizeof(T) is aligned (rows are x + c*F*sizeof(T)).
+template <int NB> struct RawVec;
+template <> struct RawVec<16> { using type = uint4; };
+template <> struct RawVec<8>  { using type = uint2; };
+template <> struct RawVec<4>  { using type = unsigned; };
+template <> struct RawVec<2>  { using type = ushort; };
+
+template <int SW, typename T, int VE>
 __global__ void spmm_csr_kernel(
     const int64_t* __restrict__ indices,
     const T* __restrict__ xl, const T* __restrict__ xr,
@@ -205,8 +214,8 @@ __global__ void spmm_csr_kernel(
     const int64_t* __restrict__ seg_e1, const uint8_t* __restrict__ seg_multi,
     int64_t n_seg, int64_t F, int64_t n_local) {
     constexpr bool BF = sizeof(T) == 2;
-    constexpr int V = 16 / sizeof(T);          // features per lane (16B loads)
-    typedef float vf4 __attribute__((ext_vector_type(4)));
+    constexpr int NB = VE * sizeof(T);
+    using Raw = typename RawVec<NB>::type;
     const int rows_per_block = blockDim.x / SW;
     // bijective XCD swizzle: blocks [0,nwg) -> xcd-contiguous chunks
     const int64_t nwg = gridDim.x;
@@ -222,77 +231,62 @@ __global__ void spmm_csr_kernel(
         const int64_t e0 = seg_e0[it], e1 = seg_e1[it];
         const bool multi = seg_multi[it];
         const float ds = dst_scale ? dst_scale[r] : 1.f;
-        for (int64_t f0 = (int64_t)sl * V; f0 < F; f0 += (int64_t)SW * V) {
-            float acc0[V], acc1[V];
+        for (int64_t f0 = (int64_t)sl * VE; f0 < F; f0 += (int64_t)SW * VE) {
+            float acc0[VE], acc1[VE];
 #pragma unroll
-            for (int k = 0; k < V; ++k) { acc0[k] = 0.f; acc1[k] = 0.f; }
-            const bool full = (f0 + V <= F);
+            for (int k = 0; k < VE; ++k) { acc0[k] = 0.f; acc1[k] = 0.f; }
             int64_t e = e0;
-            if (full) {
-                for (; e + 1 < e1; e += 2) {
-                    const int64_t c0 = NT ? __builtin_nontemporal_load(indices + e) : indices[e];
-                    const int64_t c1 = NT ? __builtin_nontemporal_load(indices + e + 1) : indices[e + 1];
-                    const float s0 = src_scale ? src_scale[c0] : 1.f;
-                    const float s1 = src_scale ? src_scale[c1] : 1.f;
-                    const T* p0 = (c0 < n_local ? xl + c0 * F : xr + (c0 - n_local) * F) + f0;
-                    const T* p1 = (c1 < n_local ? xl + c1 * F : xr + (c1 - n_local) * F) + f0;
-                    const uint4 r0 = *reinterpret_cast<const uint4*>(p0);
-                    const uint4 r1 = *reinterpret_cast<const uint4*>(p1);
-                    const T* v0 = reinterpret_cast<const T*>(&r0);
-                    const T* v1 = reinterpret_cast<const T*>(&r1);
+            for (; e + 1 < e1; e += 2) {   // VE divides F: loads always in-bounds
+                const int64_t c0 = indices[e];
+                const int64_t c1 = indices[e + 1];
+                const float s0 = src_scale ? src_scale[c0] : 1.f;
+                const float s1 = src_scale ? src_scale[c1] : 1.f;
+                const T* p0 = (c0 < n_local ? xl + c0 * F : xr + (c0 - n_local) * F) + f0;
+                const T* p1 = (c1 < n_local ? xl + c1 * F : xr + (c1 - n_local) * F) + f0;
+                const Raw r0 = *reinterpret_cast<const Raw*>(p0);
+                const Raw r1 = *reinterpret_cast<const Raw*>(p1);
+                const T* v0 = reinterpret_cast<const T*>(&r0);
+                const T* v1 = reinterpret_cast<const T*>(&r1);
 #pragma unroll
-                    for (int k = 0; k < V; ++k) {
-                        acc0[k] = fmaf(to_f32<T>(v0[k]), s0, acc0[k]);
-                        acc1[k] = fmaf(to_f32<T>(v1[k]), s1, acc1[k]);
-                    }
+                for (int k = 0; k < VE; ++k) {
+                    acc0[k] = fmaf(to_f32<T>(v0[k]), s0, acc0[k]);
+                    acc1[k] = fmaf(to_f32<T>(v1[k]), s1, acc1[k]);
                 }
             }
             for (; e < e1; ++e) {
                 const int64_t c = indices[e];
                 const float s = src_scale ? src_scale[c] : 1.f;
                 const T* xc = (c < n_local ? xl + c * F : xr + (c - n_local) * F) + f0;
+                const Raw rv = *reinterpret_cast<const Raw*>(xc);
+                const T* v = reinterpret_cast<const T*>(&rv);
 #pragma unroll
-                for (int k = 0; k < V; ++k)
-                    if (f0 + k < F)
-                        acc0[k] = fmaf(to_f32<T>(xc[k]), s, acc0[k]);
+                for (int k = 0; k < VE; ++k)
+                    acc0[k] = fmaf(to_f32<T>(v[k]), s, acc0[k]);
             }
             T* yr = y + r * F + f0;
             if (!multi) {
-                T outv[V];
+                T outv[VE];
 #pragma unroll
-                for (int k = 0; k < V; ++k) {
+                for (int k = 0; k < VE; ++k) {
                     const float o = (acc0[k] + acc1[k]) * ds;
                     if constexpr (BF) outv[k] = f32_to_bf16(o);
                     else outv[k] = o;
                 }
-                if (full) {
-                    if (NT)
-                        __builtin_nontemporal_store(*reinterpret_cast<const vf4*>(outv),
-                                                    reinterpret_cast<vf4*>(yr));
-                    else
-                        *reinterpret_cast<uint4*>(yr) = *reinterpret_cast<const uint4*>(outv);
-                } else {
-                    for (int k = 0; k < V && f0 + k < F; ++k)
-                        yr[k] = outv[k];
+                *reinterpret_cast<Raw*>(yr) = *reinterpret_cast<const Raw*>(outv);
+            } else if constexpr (BF) {
+                // long row split across segments: combine via packed atomics
+#pragma unroll
+                for (int k = 0; k < VE; k += 2) {
+                    __hip_bfloat162 v;
+                    v.x = __float2bfloat16((acc0[k] + acc1[k]) * ds);
+                    v.y = __float2bfloat16((acc0[k + 1] + acc1[k + 1]) * ds);
+                    unsafeAtomicAdd(reinterpret_cast<__hip_bfloat162*>(yr + k), v);
                 }
             } else {
-                // long row split across segments: combine via atomics
-                if constexpr (BF) {
 #pragma unroll
-                    for (int k = 0; k < V; k += 2) {
-                        if (f0 + k < F) {
-                            __hip_bfloat162 v;
-                            v.x = __float2bfloat16((acc0[k] + acc1[k]) * ds);
-                            v.y = __float2bfloat16(f0 + k + 1 < F
-                                                   ? (acc0[k + 1] + acc1[k + 1]) * ds : 0.f);
-                            unsafeAtomicAdd(reinterpret_cast<__hip_bfloat162*>(yr + k), v);
-                        }
-                    }
-                } else {
-                    for (int k = 0; k < V && f0 + k < F; ++k)
-                        atomicAdd(reinterpret_cast<float*>(yr) + k,
-                                  (acc0[k] + acc1[k]) * ds);
-                }
+                for (int k = 0; k < VE; ++k)
+                    atomicAdd(reinterpret_cast<float*>(yr) + k,
+                              (acc0[k] + acc1[k]) * ds);
             }
         }
     }
@@ -393,19 +387,19 @@ void spmm_csr(torch::Tensor indices, torch::Tensor xl,
                 zero_rows.numel(), F);
     }
     if (n_seg == 0) return;
-    // sub-wavefront width: smallest of {16,32,64} covering F with 16B lanes
-    const int vpl = bf16 ? 8 : 4;            // features per lane
+    // largest vector width dividing F (keeps every row-base load aligned)
+    int ve = bf16 ? 8 : 4;
+    while (ve > 1 && F % ve) ve >>= 1;
+    if (bf16 && ve == 1) ve = 2;              // bf16 atomics need even lanes
+    TORCH_CHECK(!bf16 || F % 2 == 0, "bf16 spmm needs even feature dim");
+    // sub-wavefront width: smallest of {16,32,64} covering F with ve/lane
     int sw = 16;
-    while (sw < 64 && sw * vpl < F) sw *= 2;
+    while (sw < 64 && (int64_t)sw * ve < F) sw *= 2;
     const int block_threads = WAVE * 4;
     const int rows_per_block = block_threads / sw;
     int64_t blocks = (n_seg + rows_per_block - 1) / rows_per_block;
     blocks = std::max<int64_t>(std::min<int64_t>(blocks, 16384), 1);
     const dim3 grid(blocks), block(block_threads);
-    static const bool nt = [] {
-        const char* v = getenv("ADAQP_SPMM_NT");
-        return v && v[0] == '1';
-    }();
     const int64_t* ind_p = indices.data_ptr<int64_t>();
     const float* ss_p = src_scale.numel() ? src_scale.data_ptr<float>() : nullptr;
     const float* ds_p = dst_scale.numel() ? dst_scale.data_ptr<float>() : nullptr;
@@ -413,31 +407,37 @@ void spmm_csr(torch::Tensor indices, torch::Tensor xl,
     const int64_t* e0_p = seg_e0.data_ptr<int64_t>();
     const int64_t* e1_p = seg_e1.data_ptr<int64_t>();
     const uint8_t* sm_p = seg_multi.data_ptr<uint8_t>();
-    auto dispatch = [&](auto sw_tag, auto nt_tag, auto t_tag) {
+    auto run = [&](auto sw_tag, auto t_tag, auto ve_tag) {
         constexpr int SWC = decltype(sw_tag)::value;
-        constexpr bool NTC = decltype(nt_tag)::value;
         using TC = typename decltype(t_tag)::type;
-        spmm_csr_kernel<SWC, NTC, TC><<<grid, block, 0, s>>>(
+        constexpr int VEC = decltype(ve_tag)::value;
+        spmm_csr_kernel<SWC, TC, VEC><<<grid, block, 0, s>>>(
             ind_p, reinterpret_cast<const TC*>(xl.data_ptr()),
             reinterpret_cast<const TC*>(xr_ptr),
             reinterpret_cast<TC*>(y.data_ptr()), ss_p, ds_p,
             sr_p, e0_p, e1_p, sm_p, n_seg, F, n_local);
     };
-    auto dis2 = [&](auto sw_tag) {
-        struct FT { using type = float; };
-        struct BT { using type = ushort; };
+    struct FT { using type = float; };
+    struct BT { using type = ushort; };
+    auto by_ve = [&](auto sw_tag) {
         if (bf16) {
-            if (nt) dispatch(sw_tag, std::true_type{}, BT{});
-            else dispatch(sw_tag, std::false_type{}, BT{});
+            switch (ve) {
+                case 8: run(sw_tag, BT{}, std::integral_constant<int,8>{}); break;
+                case 4: run(sw_tag, BT{}, std::integral_constant<int,4>{}); break;
+                default: run(sw_tag, BT{}, std::integral_constant<int,2>{}); break;
+            }
         } else {
-            if (nt) dispatch(sw_tag, std::true_type{}, FT{});
-            else dispatch(sw_tag, std::false_type{}, FT{});
+            switch (ve) {
+                case 4: run(sw_tag, FT{}, std::integral_constant<int,4>{}); break;
+                case 2: run(sw_tag, FT{}, std::integral_constant<int,2>{}); break;
+                default: run(sw_tag, FT{}, std::integral_constant<int,1>{}); break;
+            }
         }
     };
     switch (sw) {
-        case 64: dis2(std::integral_constant<int,64>{}); break;
-        case 32: dis2(std::integral_constant<int,32>{}); break;
-        default: dis2(std::integral_constant<int,16>{}); break;
+        case 64: by_ve(std::integral_constant<int,64>{}); break;
+        case 32: by_ve(std::integral_constant<int,32>{}); break;
+        default: by_ve(std::integral_constant<int,16>{}); break;
     }
 }
 

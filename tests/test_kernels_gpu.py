@@ -90,7 +90,7 @@ def test_quant_unbiased_gpu():
     assert ((mean - x).abs() <= tol).all()
 
 
-@pytest.mark.parametrize('F', [64, 100, 256])
+@pytest.mark.parametrize('F', [64, 100, 256, 602, 17])
 def test_spmm_matches_torch(F):
     from adaqp_amd.graph import random_partitioned_graph, partition_all
     from adaqp_amd.ops.kernels import SpmmView, spmm
@@ -193,7 +193,7 @@ def test_spmm_bf16_matches_fp32(etol=0.02):
     torch.manual_seed(6)
     g = random_partitioned_graph(800, 8000, 4, 3, 1, seed=12)
     lg = partition_all(g, 1)[0]
-    for F in (100, 256):
+    for F in (100, 256, 602):
         x = torch.randn(lg.num_nodes, F, device='cuda')
         src = torch.rand(lg.num_nodes, device='cuda') + 0.5
         dst = torch.rand(lg.num_inner, device='cuda') + 0.5
