@@ -154,3 +154,17 @@ class LocalGraph:
                 setattr(self, name, t.to(device))
         self.send_idx = {p: v.to(device) for p, v in self.send_idx.items()}
         return self
+
+
+def pad_feat_dim(graph: 'LocalGraph', multiple: int = 8) -> int:
+    """Zero-pad the feature dim to a multiple of ``multiple`` so every
+    SpMM row-base load is wide-vector aligned (602 -> 608 etc.). Zero
+    columns are inert: their weights receive zero gradient. Returns the
+    padded dim."""
+    if graph.feats is None:
+        return 0
+    F = graph.feats.shape[1]
+    pad = (-F) % multiple
+    if pad:
+        graph.feats = torch.nn.functional.pad(graph.feats, (0, pad))
+    return F + pad

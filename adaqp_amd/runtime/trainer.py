@@ -18,8 +18,8 @@ import torch
 import yaml
 
 from ..comm.communicator import Communicator
-from ..graph import (DATASET_SHAPES, load_partition, partition_all,
-                     save_partitions, synth_graph)
+from ..graph import (DATASET_SHAPES, load_partition, pad_feat_dim,
+                     partition_all, save_partitions, synth_graph)
 from ..helpers import AssignScheme, DistGNNType, RunMode
 from ..models import DistGCN, DistSAGE
 from ..assigner import Assigner
@@ -106,7 +106,8 @@ class Trainer:
     def _set_engine(self):
         m = self.cfg['model']
         shape = DATASET_SHAPES[self.args.dataset]
-        self.feat_dim, self.num_classes, self.multilabel = shape[2], shape[3], shape[4]
+        self.num_classes, self.multilabel = shape[3], shape[4]
+        self.feat_dim = pad_feat_dim(self.graph, 8)
         L = m['num_layers']
         msg_dims = [self.feat_dim] + [m['hidden_dim']] * (L - 1)
         self.engine = GraphEngine(self.graph, self.mode, self.model_type,

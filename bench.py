@@ -91,7 +91,9 @@ def main():
     comm = Communicator(backend=None if use_gpu else 'gloo')
     lg = prepare_partition(args, rank, world)
 
-    _, _, feat_dim, num_classes, multilabel = DATASET_SHAPES[args.dataset]
+    from adaqp_amd.graph import pad_feat_dim
+    _, _, _, num_classes, multilabel = DATASET_SHAPES[args.dataset]
+    feat_dim = pad_feat_dim(lg, 8)
     mode = RunMode(args.mode)
     mtype = DistGNNType.DistGCN if args.model == 'gcn' else DistGNNType.DistSAGE
     msg_dims = [feat_dim] + [args.hidden] * (args.layers - 1)
