@@ -145,12 +145,14 @@ class Trainer:
                                           lr=self.cfg['runtime']['lr'])
 
     # ------------------------------------------------------------------
-    def train(self) -> Dict[str, float]:
+    def train(self, start_epoch: int = 0) -> Dict[str, float]:
         rt = self.cfg['runtime']
         gc = global_train_count(self.engine)
         quant = self.mode.bit_type.name == 'QUANT'
         adaptive = quant and self.assigner.scheme == AssignScheme.ADAPTIVE
-        for epoch in range(rt['num_epochs']):
+        ckpt_every = getattr(self.args, 'ckpt_every', None)
+        ckpt_path = getattr(self.args, 'ckpt_path', None)
+        for epoch in range(start_epoch, rt['num_epochs']):
             if adaptive and epoch > 0 and epoch % self.assign_cycle == 0:
                 self.assigner.reassign()
             t0 = time.perf_counter()
@@ -170,7 +172,27 @@ class Trainer:
                     f'train {metrics["train"]:.4f} val {metrics["val"]:.4f} '
                     f'test {metrics["test"]:.4f} '
                     f'epoch_time {self.epoch_times[-1]*1e3:.1f}ms')
+            if ckpt_every and ckpt_path and (epoch + 1) % ckpt_every == 0:
+                self.save_checkpoint(ckpt_path, epoch + 1)
         return self.recorder.best()
+
+    # ---- model checkpointing (absent in the reference; completes the
+    # framework's persistence story alongside the partition artifacts) ----
+    def save_checkpoint(self, path: str, epoch: int) -> None:
+        if self.comm.rank == 0:
+            os.makedirs(os.path.dirname(path) or '.', exist_ok=True)
+            torch.save({'epoch': epoch,
+                        'model': self.model.state_dict(),
+                        'optimizer': self.optimizer.state_dict()}, path)
+        self.comm.barrier()
+
+    def load_checkpoint(self, path: str) -> int:
+        state = torch.load(path, map_location=self.comm.device,
+                           weights_only=False)
+        self.model.load_state_dict(state['model'])
+        self.optimizer.load_state_dict(state['optimizer'])
+        self.comm.sync_model_params(self.model)
+        return int(state['epoch'])
 
     def save(self, root: str = 'exp'):
         args = self.args

@@ -37,6 +37,9 @@ def main():
     p.add_argument('--log_steps', type=int, default=None)
     p.add_argument('--seed', type=int, default=None)
     p.add_argument('--dtype', type=str, default='fp32', choices=['fp32', 'bf16'])
+    p.add_argument('--ckpt_path', type=str, default=None)
+    p.add_argument('--ckpt_every', type=int, default=None)
+    p.add_argument('--resume', action='store_true')
     p.add_argument('--time_breakdown', action='store_true',
                    help='enable per-epoch comm/quant/agg span timing '
                         '(adds sync fences; use rocprofv3 for kernel evidence)')
@@ -56,7 +59,11 @@ def main():
                          'launch with --nproc-per-node equal to num_parts')
 
     trainer = Trainer(args)
-    best = trainer.train()
+    start = 0
+    if args.resume and args.ckpt_path and os.path.exists(args.ckpt_path):
+        start = trainer.load_checkpoint(args.ckpt_path)
+        print(f'resumed from {args.ckpt_path} at epoch {start}')
+    best = trainer.train(start_epoch=start)
     trainer.save()
     if trainer.comm.rank == 0:
         print(f'best: epoch {best["epoch"]} val {best["val"]:.4f} '

@@ -42,3 +42,26 @@ def test_main_cli_runs(tmp_path):
     exp = tmp_path / 'exp' / 'reddit' / '1part' / 'sage'
     assert (exp / 'metrics').is_dir()
     assert (exp / 'time').is_dir()
+
+
+def test_checkpoint_resume(tmp_path):
+    env = dict(os.environ, MASTER_PORT='29533')
+    subprocess.run(
+        [sys.executable, 'graph_partition.py', '--dataset', 'reddit',
+         '--partition_size', '1', '--scale', '0.002',
+         '--partition_dir', str(tmp_path / 'parts')],
+        cwd=REPO, check=True, capture_output=True, timeout=300)
+    ck = str(tmp_path / 'ckpt.pt')
+    base = [sys.executable, os.path.join(REPO, 'main.py'), '--dataset',
+            'reddit', '--model_name', 'gcn', '--mode', 'Vanilla',
+            '--log_steps', '1', '--partition_dir', str(tmp_path / 'parts'),
+            '--ckpt_path', ck, '--ckpt_every', '2']
+    out = subprocess.run(base + ['--num_epochs', '4'], cwd=str(tmp_path),
+                         env=env, capture_output=True, text=True, timeout=600)
+    assert out.returncode == 0, out.stderr[-1500:]
+    assert os.path.exists(ck)
+    out = subprocess.run(base + ['--num_epochs', '6', '--resume'],
+                         cwd=str(tmp_path), env=env, capture_output=True,
+                         text=True, timeout=600)
+    assert out.returncode == 0, out.stderr[-1500:]
+    assert 'resumed from' in out.stdout
