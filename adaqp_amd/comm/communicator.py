@@ -90,7 +90,19 @@ class Communicator:
                      out_splits: Sequence[int], in_splits: Sequence[int],
                      async_op: bool = False):
         """Single fused variable-size all-to-all. On RCCL this is grouped
-        xGMI send/recv on the current stream; on gloo, alltoallv."""
+        xGMI send/recv on the current stream; on gloo, alltoallv.
+
+        gloo + CUDA tensors (debug/testing: several ranks sharing one
+        GPU, or forcing the reference's CPU-staged transport) stages
+        through host memory like the reference's pinned-CPU path
+        (``comm.py:173-189``)."""
+        if out.is_cuda and 'nccl' not in self.backend:
+            inp_c = inp.cpu()
+            out_c = torch.empty(out.shape, dtype=out.dtype)
+            dist.all_to_all_single(out_c, inp_c, list(out_splits),
+                                   list(in_splits))
+            out.copy_(out_c, non_blocking=False)
+            return None
         return dist.all_to_all_single(out, inp, list(out_splits),
                                       list(in_splits), async_op=async_op)
 
