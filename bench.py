@@ -18,6 +18,15 @@ import os
 import sys
 import time
 
+# pre-tuned hipBLASLt GEMM selections (bf16, gfx950) — read-only lookup;
+# untuned shapes fall back to the default backend. Must be set before torch.
+_REPO = os.path.dirname(os.path.abspath(__file__))
+_TUNED = os.path.join(_REPO, 'adaqp_amd', 'tuned', 'tunableop.csv')
+if os.path.exists(os.path.join(_REPO, 'adaqp_amd', 'tuned', 'tunableop0.csv')):
+    os.environ.setdefault('PYTORCH_TUNABLEOP_ENABLED', '1')
+    os.environ.setdefault('PYTORCH_TUNABLEOP_TUNING', '0')
+    os.environ.setdefault('PYTORCH_TUNABLEOP_FILENAME', _TUNED)
+
 import torch
 
 
