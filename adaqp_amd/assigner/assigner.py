@@ -31,9 +31,9 @@ import numpy as np
 import torch
 from torch import Tensor
 
-from ..comm.buffers import BITS_SET, bytes_per_node, uniform_bits
+from ..comm.buffers import BITS_SET, bytes_per_node
 from ..comm.communicator import Communicator
-from ..helpers import AssignScheme, DistGNNType, PropagationMode
+from ..helpers import AssignScheme, DistGNNType
 from .profile import fit_cost_models
 
 BITS_COST = {b: 1.0 / (2.0 ** b - 1) ** 2 for b in BITS_SET}

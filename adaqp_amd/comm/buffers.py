@@ -23,8 +23,8 @@ after every re-assignment — replacing the reference's
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
-from typing import Dict, List, Optional, Sequence, Tuple
+from dataclasses import dataclass
+from typing import Dict, List, Optional
 
 import torch
 from torch import Tensor

@@ -23,7 +23,7 @@ import torch
 from torch import Tensor
 
 from ..graph import LocalGraph
-from ..helpers import BitType, DistGNNType, RunMode
+from ..helpers import DistGNNType, RunMode
 from ..comm.buffers import KeyPlan, build_key_plan, uniform_bits
 from ..comm.communicator import Communicator
 from .timer import Timer

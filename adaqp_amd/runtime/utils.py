@@ -8,7 +8,7 @@ correctly scaled — same convention here.
 """
 from __future__ import annotations
 
-from typing import Dict, Tuple
+from typing import Dict
 
 import torch
 import torch.nn.functional as F
