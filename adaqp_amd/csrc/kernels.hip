@@ -293,6 +293,96 @@ __global__ void spmm_csr_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// fused dual GEMM (bf16, MFMA): out[M,N] = A1@W1 + A2@W2 + bias
+//
+// The SAGE 'mean' layer computes fc_self(x) + fc_neigh(h) + biases — two
+// rocBLAS GEMMs plus an elementwise add, writing/reading the [M,N] output
+// three times. This kernel produces it in ONE pass: per 64-row block, the
+// TRANSPOSED weights (Wt = W.T, [N,K]) are staged through LDS (shared by
+// the block's 4 waves; W1 first, then W2), and each wave accumulates its
+// 16xN rows with v_mfma_f32_16x16x32_bf16. fp32 accumulate, bf16 I/O.
+// Fragment layouts per guide §3: A row = lane&15, k = (lane>>4)*8+j;
+// B col = lane&15, same k; C/D col = lane&15, row = (lane>>4)*4+reg.
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define FDG_MAXNT 16      // N <= 256
+#define FDG_PAD 8         // LDS row pad (elements) to break bank conflicts
+
+__device__ __forceinline__ bf16x8 load_bf16x8(const ushort* p) {
+    uint4 raw = *reinterpret_cast<const uint4*>(p);
+    return *reinterpret_cast<const bf16x8*>(&raw);
+}
+
+__global__ void __launch_bounds__(256)
+fused_dual_gemm_bf16_kernel(
+    const ushort* __restrict__ a1, const ushort* __restrict__ a2,
+    const ushort* __restrict__ w1t, const ushort* __restrict__ w2t,
+    const ushort* __restrict__ bias,     // [N] bf16 (b1+b2), may be null
+    ushort* __restrict__ out,
+    int64_t M, int64_t N, int64_t K1, int64_t K2) {
+    extern __shared__ __attribute__((aligned(16))) ushort lds[];  // [N][Kmax+PAD]
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int64_t row0 = (int64_t)blockIdx.x * 64 + wid * 16;
+    const int arow = lane & 15;          // A row / B col within a 16-tile
+    const int kgrp = lane >> 4;          // k sub-block (8 elems each)
+    const int NT = (int)(N / 16);
+
+    f32x4 acc[FDG_MAXNT];
+#pragma unroll
+    for (int nt = 0; nt < FDG_MAXNT; ++nt) acc[nt] = {0.f, 0.f, 0.f, 0.f};
+
+    for (int pass = 0; pass < 2; ++pass) {
+        const ushort* A = pass ? a2 : a1;
+        const ushort* Wt = pass ? w2t : w1t;
+        const int64_t K = pass ? K2 : K1;
+        const int64_t ldl = K + FDG_PAD;
+        // stage Wt[N][K] -> LDS[N][K+PAD] (whole block cooperates)
+        __syncthreads();
+        for (int64_t i = threadIdx.x * 8; i < N * K; i += (int64_t)blockDim.x * 8) {
+            const int64_t r = i / K, c = i % K;      // K % 8 == 0
+            *reinterpret_cast<uint4*>(&lds[r * ldl + c]) =
+                *reinterpret_cast<const uint4*>(&Wt[i]);
+        }
+        __syncthreads();
+        const int64_t arow_g = row0 + arow;
+        const ushort* arow_p = A + arow_g * K;
+        for (int64_t k0 = 0; k0 < K; k0 += 32) {
+            const int64_t k = k0 + kgrp * 8;
+            bf16x8 af = {};
+            if (arow_g < M && k < K)               // K % 8 == 0: whole sub-block
+                af = load_bf16x8(arow_p + k);
+#pragma unroll
+            for (int nt = 0; nt < FDG_MAXNT; ++nt) {
+                if (nt >= NT) break;
+                bf16x8 bf = {};
+                if (k < K)
+                    bf = *reinterpret_cast<const bf16x8*>(
+                        &lds[(nt * 16 + arow) * ldl + k]);
+                acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    af, bf, acc[nt], 0, 0, 0);
+            }
+        }
+    }
+
+    // epilogue: bias + bf16 store. C/D: col = lane&15, row = kgrp*4 + q
+#pragma unroll
+    for (int nt = 0; nt < FDG_MAXNT; ++nt) {
+        if (nt >= NT) break;
+        const int col = nt * 16 + arow;
+        const float bv = bias ? bf16_to_f32(bias[col]) : 0.f;
+#pragma unroll
+        for (int q = 0; q < 4; ++q) {
+            const int64_t r = row0 + kgrp * 4 + q;
+            if (r < M)
+                out[r * N + col] = f32_to_bf16(acc[nt][q] + bv);
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
 // host launchers
 // ---------------------------------------------------------------------------
 static inline hipStream_t cur_stream() {
@@ -441,6 +531,34 @@ void spmm_csr(torch::Tensor indices, torch::Tensor xl,
     }
 }
 
+
+void fused_dual_gemm_bf16(torch::Tensor a1, torch::Tensor a2,
+                          torch::Tensor w1t, torch::Tensor w2t,
+                          torch::Tensor bias, torch::Tensor out) {
+    CHECK_DEV(a1); CHECK_CONTIG(a1); CHECK_DEV(a2); CHECK_CONTIG(a2);
+    CHECK_DEV(w1t); CHECK_CONTIG(w1t); CHECK_DEV(w2t); CHECK_CONTIG(w2t);
+    CHECK_DEV(out); CHECK_CONTIG(out);
+    TORCH_CHECK(a1.scalar_type() == torch::kBFloat16, "bf16 only");
+    const int64_t M = a1.size(0), K1 = a1.size(1), K2 = a2.size(1);
+    const int64_t N = out.size(1);
+    TORCH_CHECK(a2.size(0) == M && out.size(0) == M, "M mismatch");
+    TORCH_CHECK(w1t.size(0) == N && w1t.size(1) == K1, "w1t must be [N,K1]");
+    TORCH_CHECK(w2t.size(0) == N && w2t.size(1) == K2, "w2t must be [N,K2]");
+    TORCH_CHECK(N % 16 == 0 && N <= 256, "N must be mult of 16, <=256");
+    TORCH_CHECK(K1 % 8 == 0 && K2 % 8 == 0, "K must be mult of 8");
+    const int64_t kmax = std::max(K1, K2);
+    const size_t lds_bytes = (size_t)N * (kmax + 8) * 2;
+    TORCH_CHECK(lds_bytes <= 160 * 1024, "weights too large for LDS staging");
+    const dim3 block(256), grid((M + 63) / 64);
+    fused_dual_gemm_bf16_kernel<<<grid, block, lds_bytes, cur_stream()>>>(
+        reinterpret_cast<const ushort*>(a1.data_ptr()),
+        reinterpret_cast<const ushort*>(a2.data_ptr()),
+        reinterpret_cast<const ushort*>(w1t.data_ptr()),
+        reinterpret_cast<const ushort*>(w2t.data_ptr()),
+        bias.numel() ? reinterpret_cast<const ushort*>(bias.data_ptr()) : nullptr,
+        reinterpret_cast<ushort*>(out.data_ptr()), M, N, K1, K2);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("quant_pack", &quant_pack,
           "fused minmax+stochastic-quantize+bitpack (CDNA4)");
@@ -448,4 +566,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused dequantize+scatter (CDNA4)");
     m.def("spmm_csr", &spmm_csr,
           "CSR SpMM with fused degree normalization (CDNA4)");
+    m.def("fused_dual_gemm_bf16", &fused_dual_gemm_bf16,
+          "out = A1@W1 + A2@W2 + bias, one-pass MFMA (CDNA4)");
 }

@@ -258,3 +258,41 @@ def test_e2e_train_step_bf16():
         l = float(train_epoch(engine, model, opt, gc, False))
     assert torch.isfinite(torch.tensor(l)) and l < l0
     Communicator.shutdown()
+
+
+def test_fused_dual_gemm_bf16():
+    """Hand-written MFMA kernel vs torch reference. Uses ASYMMETRIC
+    weights (guide G9: symmetric B hides transposed C-writes)."""
+    C = _native()
+    torch.manual_seed(9)
+    for (M, K1, K2, N) in [(1000, 104, 256, 256), (64, 32, 32, 64),
+                           (77, 8, 16, 48)]:
+        a1 = torch.randn(M, K1, device='cuda').bfloat16()
+        a2 = torch.randn(M, K2, device='cuda').bfloat16()
+        w1 = torch.randn(K1, N, device='cuda').bfloat16()
+        w2 = torch.randn(K2, N, device='cuda').bfloat16()
+        bias = torch.randn(N, device='cuda').bfloat16()
+        out = torch.empty(M, N, device='cuda', dtype=torch.bfloat16)
+        C.fused_dual_gemm_bf16(a1, a2, w1.t().contiguous(),
+                               w2.t().contiguous(), bias, out)
+        ref = (a1.float() @ w1.float() + a2.float() @ w2.float()
+               + bias.float())
+        scale = ref.abs().max().clamp(min=1.0)
+        err = (out.float() - ref).abs().max() / scale
+        assert err < 0.02, (M, K1, K2, N, float(err))
+    # transpose detection: A = one-hot rows, W asymmetric
+    M, K, N = 32, 32, 32
+    a1 = torch.zeros(M, K, device='cuda')
+    a1[torch.arange(M), torch.arange(M) % K] = 1.0
+    a1 = a1.bfloat16()
+    a2 = torch.zeros(M, K, device='cuda').bfloat16()
+    w1 = (torch.arange(K, device='cuda')[:, None] * 100.0
+          + torch.arange(N, device='cuda')[None, :]).bfloat16()
+    w2 = torch.zeros(K, N, device='cuda').bfloat16()
+    out = torch.empty(M, N, device='cuda', dtype=torch.bfloat16)
+    C.fused_dual_gemm_bf16(a1, a2, w1.t().contiguous(), w2.t().contiguous(),
+                           torch.empty(0, device='cuda', dtype=torch.bfloat16),
+                           out)
+    ref = a1.float() @ w1.float()
+    assert torch.allclose(out.float(), ref, atol=2.0), \
+        (out.float() - ref).abs().max()
