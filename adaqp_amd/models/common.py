@@ -73,3 +73,54 @@ class FastLinear(nn.Module):
         if b is None:
             return x @ w
         return _LinearBiasFn.apply(x, w, b)
+
+
+class _FusedDualLinearFn(torch.autograd.Function):
+    """out = x @ w1 + h @ w2 + (b1 + b2) via the one-pass MFMA kernel
+    (csrc fused_dual_gemm_bf16); backward through plain GEMMs."""
+
+    @staticmethod
+    def forward(ctx, x: Tensor, h: Tensor, w1: Tensor, w2: Tensor,
+                b1: Tensor, b2: Tensor) -> Tensor:
+        from ..ops.kernels import native
+        ctx.save_for_backward(x, h, w1, w2)
+        M, N = x.shape[0], w1.shape[1]
+        out = torch.empty(M, N, dtype=x.dtype, device=x.device)
+        bias = ((b1 + b2).to(x.dtype) if b1 is not None
+                else torch.empty(0, dtype=x.dtype, device=x.device))
+        native().fused_dual_gemm_bf16(x.contiguous(), h.contiguous(),
+                                      w1.t().contiguous(), w2.t().contiguous(),
+                                      bias, out)
+        return out
+
+    @staticmethod
+    def backward(ctx, g: Tensor):
+        x, h, w1, w2 = ctx.saved_tensors
+        g = g.contiguous()
+        gx = g @ w1.t() if ctx.needs_input_grad[0] else None
+        gh = g @ w2.t() if ctx.needs_input_grad[1] else None
+        gw1 = x.t() @ g if ctx.needs_input_grad[2] else None
+        gw2 = h.t() @ g if ctx.needs_input_grad[3] else None
+        gb = None
+        if ctx.needs_input_grad[4]:
+            gb = (_ones_row(g.shape[0], g.device, g.dtype) @ g).reshape(-1)
+        return gx, gh, gw1, gw2, gb, gb
+
+
+def fused_dual_linear_ok(x: Tensor, h: Tensor, n_out: int) -> bool:
+    from ..ops.kernels import has_native
+    return (x.is_cuda and x.dtype == torch.bfloat16
+            and h.dtype == torch.bfloat16
+            and n_out % 16 == 0 and n_out <= 256
+            and x.shape[1] % 8 == 0 and h.shape[1] % 8 == 0
+            and has_native())
+
+
+def fused_dual_linear(x, h, lin1: 'FastLinear', lin2: 'FastLinear') -> Tensor:
+    """lin1(x) + lin2(h) in one MFMA pass (bf16 CUDA); casts mirror
+    FastLinear.forward so fp32 master weights get fp32 grads."""
+    dt = torch.bfloat16
+    return _FusedDualLinearFn.apply(
+        x.to(dt), h.to(dt), lin1.weight.to(dt), lin2.weight.to(dt),
+        lin1.bias.to(dt) if lin1.bias is not None else None,
+        lin2.bias.to(dt) if lin2.bias is not None else None)

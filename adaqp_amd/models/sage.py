@@ -10,7 +10,7 @@ import torch.nn as nn
 from torch import Tensor
 
 from ..ops.dist_agg import dist_aggregate
-from .common import FastLinear
+from .common import FastLinear, fused_dual_linear, fused_dual_linear_ok
 
 
 class DistSAGEConv(nn.Module):
@@ -32,7 +32,14 @@ class DistSAGEConv(nn.Module):
     def forward(self, engine, x: Tensor) -> Tensor:
         h_neigh = dist_aggregate(x, engine, self.layer, self.training)
         if self.aggregator_type == 'mean':
-            return self.fc_self(x[:engine.graph.num_inner]) + self.fc_neigh(h_neigh)
+            x_self = x[:engine.graph.num_inner]
+            if (engine.compute_dtype == torch.bfloat16
+                    and fused_dual_linear_ok(h_neigh, h_neigh,
+                                             self.fc_neigh.weight.shape[1])
+                    and x_self.shape[1] % 8 == 0):
+                return fused_dual_linear(x_self, h_neigh,
+                                         self.fc_self, self.fc_neigh)
+            return self.fc_self(x_self) + self.fc_neigh(h_neigh)
         return self.fc_neigh(h_neigh)
 
 
