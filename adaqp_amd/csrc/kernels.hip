@@ -322,7 +322,9 @@ fused_dual_gemm_bf16_kernel(
     const ushort* __restrict__ bias,     // [N] bf16 (b1+b2), may be null
     ushort* __restrict__ out,
     int64_t M, int64_t N, int64_t K1, int64_t K2) {
-    extern __shared__ __attribute__((aligned(16))) ushort lds[];  // [N][Kmax+PAD]
+    // Wt fragments are read straight from global: the whole weight set
+    // (<=256KB) is L2-resident and re-read by every block; LDS staging
+    // was measured SLOWER (135KB/block -> 1 block/CU, no latency hiding).
     const int wid = threadIdx.x / WAVE;
     const int lane = threadIdx.x & (WAVE - 1);
     const int64_t row0 = (int64_t)blockIdx.x * 64 + wid * 16;
@@ -338,31 +340,26 @@ fused_dual_gemm_bf16_kernel(
         const ushort* A = pass ? a2 : a1;
         const ushort* Wt = pass ? w2t : w1t;
         const int64_t K = pass ? K2 : K1;
-        const int64_t ldl = K + FDG_PAD;
-        // stage Wt[N][K] -> LDS[N][K+PAD] (whole block cooperates)
-        __syncthreads();
-        for (int64_t i = threadIdx.x * 8; i < N * K; i += (int64_t)blockDim.x * 8) {
-            const int64_t r = i / K, c = i % K;      // K % 8 == 0
-            *reinterpret_cast<uint4*>(&lds[r * ldl + c]) =
-                *reinterpret_cast<const uint4*>(&Wt[i]);
-        }
-        __syncthreads();
         const int64_t arow_g = row0 + arow;
         const ushort* arow_p = A + arow_g * K;
+        const ushort* wt_p = Wt + arow * K;   // row (=out col) within 16-tile
         for (int64_t k0 = 0; k0 < K; k0 += 32) {
             const int64_t k = k0 + kgrp * 8;
             bf16x8 af = {};
             if (arow_g < M && k < K)               // K % 8 == 0: whole sub-block
                 af = load_bf16x8(arow_p + k);
+            bf16x8 bfr[FDG_MAXNT];
 #pragma unroll
             for (int nt = 0; nt < FDG_MAXNT; ++nt) {
                 if (nt >= NT) break;
-                bf16x8 bf = {};
-                if (k < K)
-                    bf = *reinterpret_cast<const bf16x8*>(
-                        &lds[(nt * 16 + arow) * ldl + k]);
+                bfr[nt] = (k < K) ? load_bf16x8(wt_p + (int64_t)nt * 16 * K + k)
+                                  : bf16x8{};
+            }
+#pragma unroll
+            for (int nt = 0; nt < FDG_MAXNT; ++nt) {
+                if (nt >= NT) break;
                 acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    af, bf, acc[nt], 0, 0, 0);
+                    af, bfr[nt], acc[nt], 0, 0, 0);
             }
         }
     }
@@ -546,11 +543,8 @@ void fused_dual_gemm_bf16(torch::Tensor a1, torch::Tensor a2,
     TORCH_CHECK(w2t.size(0) == N && w2t.size(1) == K2, "w2t must be [N,K2]");
     TORCH_CHECK(N % 16 == 0 && N <= 256, "N must be mult of 16, <=256");
     TORCH_CHECK(K1 % 8 == 0 && K2 % 8 == 0, "K must be mult of 8");
-    const int64_t kmax = std::max(K1, K2);
-    const size_t lds_bytes = (size_t)N * (kmax + 8) * 2;
-    TORCH_CHECK(lds_bytes <= 160 * 1024, "weights too large for LDS staging");
     const dim3 block(256), grid((M + 63) / 64);
-    fused_dual_gemm_bf16_kernel<<<grid, block, lds_bytes, cur_stream()>>>(
+    fused_dual_gemm_bf16_kernel<<<grid, block, 0, cur_stream()>>>(
         reinterpret_cast<const ushort*>(a1.data_ptr()),
         reinterpret_cast<const ushort*>(a2.data_ptr()),
         reinterpret_cast<const ushort*>(w1t.data_ptr()),
