@@ -108,6 +108,14 @@ class _FusedDualLinearFn(torch.autograd.Function):
 
 
 def fused_dual_linear_ok(x: Tensor, h: Tensor, n_out: int) -> bool:
+    """Opt-in (ADAQP_FUSED_SAGE=1): measured SLOWER than two TUNED
+    hipBLASLt GEMMs + add on the products shape (152-177 vs 132 ms/epoch
+    — profiles/r01_NOTES.md); hipBLASLt's pipelined schedules win this
+    bandwidth-bound tall-skinny shape. Kept as a tested reference
+    implementation of a fused MFMA epilogue."""
+    import os
+    if os.environ.get('ADAQP_FUSED_SAGE') != '1':
+        return False
     from ..ops.kernels import has_native
     return (x.is_cuda and x.dtype == torch.bfloat16
             and h.dtype == torch.bfloat16
