@@ -67,22 +67,26 @@ def get_dist_env(args):
 
 
 def prepare_partition(args, rank, world):
-    """Rank 0 generates+partitions the synthetic graph once; all load."""
-    from adaqp_amd.graph import synth_graph, partition_all, save_partitions, load_partition
-    import torch.distributed as dist
+    """Every rank generates the SAME global graph (deterministic seed)
+    and builds only ITS OWN partition — fully parallel setup, no rank-0
+    serialization, no barrier imbalance. A pre-built partition cache
+    (graph_partition.py) is used when present."""
+    from adaqp_amd.graph import (synth_graph, build_local_graph,
+                                 range_assignment, global_degrees,
+                                 load_partition)
     tag = f'{args.dataset}_s{args.scale}'
     d = os.path.join(args.part_dir, tag, f'{world}part')
-    marker = os.path.join(d, f'{tag}.json')
-    if rank == 0 and not os.path.exists(marker):
-        t0 = time.time()
-        g = synth_graph(args.dataset, world, seed=17, scale=args.scale)
-        parts = partition_all(g, world, method='range')
-        save_partitions(parts, args.part_dir, tag)
+    if os.path.exists(os.path.join(d, f'{tag}.json')):
+        return load_partition(args.part_dir, tag, world, rank)
+    t0 = time.time()
+    g = synth_graph(args.dataset, world, seed=17, scale=args.scale)
+    assign = range_assignment(g.num_nodes, world)
+    in_deg, out_deg = global_degrees(g)
+    lg = build_local_graph(g, assign, rank, world, in_deg, out_deg)
+    if rank == 0:
         print(f'# partition prep {time.time()-t0:.1f}s: '
               f'{g.num_nodes} nodes {g.num_edges} edges', file=sys.stderr)
-    if world > 1:
-        dist.barrier()
-    return load_partition(args.part_dir, tag, world, rank)
+    return lg
 
 
 def main():
