@@ -73,3 +73,59 @@ def test_two_ranks_one_gpu(mode):
     for rank, loss, acc in res:
         assert torch.isfinite(torch.tensor(loss))
         assert acc > 0.4, f'mode {mode}: test acc {acc} — not learning'
+
+
+def _parity_worker(rank, world, port, q):
+    os.environ.update(MASTER_ADDR='127.0.0.1', MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK='0')
+    import torch
+    from adaqp_amd.comm import Communicator
+    from adaqp_amd.runtime import GraphEngine
+    from adaqp_amd.models import DistGCN
+    from adaqp_amd.helpers import RunMode, DistGNNType
+    from adaqp_amd.graph import random_partitioned_graph, partition_all
+    comm = Communicator(backend='gloo')
+    try:
+        dev = torch.device('cuda:0')
+        torch.manual_seed(5)
+        g = random_partitioned_graph(500, 5000, 32, 5, world, seed=9,
+                                     cut_frac=0.3)
+        lg = partition_all(g, world)[rank]
+        torch.manual_seed(77)
+        model = DistGCN(32, 32, 5, num_layers=3, dropout=0.0).to(dev)
+        comm.sync_model_params(model)
+        model.eval()
+        feats = lg.feats.to(dev)
+        outs = {}
+        for mode in ('Vanilla', 'AdaQP-p'):
+            engine = GraphEngine(lg, RunMode(mode), DistGNNType.DistGCN,
+                                 msg_dims=[32, 32, 32], device=dev)
+            with torch.no_grad():
+                outs[mode] = model(engine, feats)
+        err = (outs['Vanilla'] - outs['AdaQP-p']).abs().max().item()
+        q.put((rank, err))
+    finally:
+        Communicator.shutdown()
+
+
+def test_overlap_path_exact_parity():
+    """The stream-overlapped decomposed path must produce bit-near
+    results identical to the full path (ordering/event regression)."""
+    ctx = mp.get_context('spawn')
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_parity_worker, args=(r, P, 29549, q))
+             for r in range(P)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(300)
+        if p.is_alive():
+            p.terminate()
+            raise AssertionError('hung')
+        assert p.exitcode == 0
+    n = 0
+    while not q.empty():
+        _, err = q.get()
+        assert err < 1e-5, f'overlap path diverges from full path: {err}'
+        n += 1
+    assert n == P
