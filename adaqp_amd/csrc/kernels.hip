@@ -14,8 +14,10 @@
 //    64-wide wavefront (the reference packs along the node axis).
 //  * RNG is a stateless counter hash (triple32) of (seed, node, feature) —
 //    no curand state; bit-compatible with the CPU oracle in ops/quant.py.
-//  * spmm_csr: one 64-lane wavefront per destination row, lane l owning
-//    features [4l, 4l+4) (float4), with fused src/dst degree scaling.
+//  * spmm_csr: one SUB-wavefront per destination-row segment (hub rows are
+//    split and atomically combined), dtype-templated (fp32 / bf16 with fp32
+//    accumulate), vector width chosen so row-base loads stay aligned, fused
+//    src/dst degree scaling, XCD-aware block swizzle.
 //
 // Built for gfx950 only. No CUDA compatibility path.
 #include <hip/hip_runtime.h>
@@ -308,7 +310,6 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 #define FDG_MAXNT 16      // N <= 256
-#define FDG_PAD 8         // LDS row pad (elements) to break bank conflicts
 
 __device__ __forceinline__ bf16x8 load_bf16x8(const ushort* p) {
     uint4 raw = *reinterpret_cast<const uint4*>(p);
