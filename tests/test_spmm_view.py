@@ -1,0 +1,71 @@
+"""SpmmView segmentation invariants (the GPU kernel's work-item layout)."""
+import torch
+
+from adaqp_amd.ops.kernels import SpmmView, SEG_EDGES
+from adaqp_amd.runtime.timer import Timer
+from adaqp_amd.runtime.recorder import Recorder
+
+
+def _random_csr(rows, max_deg, seed, hub=None):
+    g = torch.Generator().manual_seed(seed)
+    deg = torch.randint(0, max_deg, (rows,), generator=g)
+    if hub is not None:
+        deg[0] = hub
+    indptr = torch.zeros(rows + 1, dtype=torch.int64)
+    indptr[1:] = torch.cumsum(deg, 0)
+    indices = torch.randint(0, rows, (int(indptr[-1]),), generator=g)
+    return indptr, indices
+
+
+def test_segments_cover_all_edges_exactly():
+    for seed in range(3):
+        rows = 200
+        indptr, indices = _random_csr(rows, 30, seed, hub=5 * SEG_EDGES + 3)
+        v = SpmmView(indptr, indices, 0, rows)
+        # every edge covered exactly once, in order, within its row
+        covered = torch.zeros(int(indptr[-1]), dtype=torch.int32)
+        for i in range(v.seg_row.numel()):
+            r = int(v.seg_row[i])
+            e0, e1 = int(v.seg_e0[i]), int(v.seg_e1[i])
+            assert indptr[r] <= e0 <= e1 <= indptr[r + 1]
+            assert e1 - e0 <= SEG_EDGES
+            covered[e0:e1] += 1
+        assert (covered == 1).all()
+        # multi flag consistent with zero_rows
+        multi_rows = set(v.zero_rows.tolist())
+        for i in range(v.seg_row.numel()):
+            assert bool(v.seg_multi[i]) == (int(v.seg_row[i]) in multi_rows)
+        # rows with 0 edges still get one (empty) segment -> output written
+        zero_deg = (indptr[1:] == indptr[:-1]).sum()
+        assert v.seg_row.numel() >= rows
+        assert v.seg_row.numel() == rows + sum(
+            max((int(indptr[r + 1] - indptr[r]) + SEG_EDGES - 1) // SEG_EDGES, 1) - 1
+            for r in range(rows))
+
+
+def test_timer_rollup():
+    t = Timer(enabled=True, cuda=False)
+    with t.record('forward0_exchange'):
+        pass
+    with t.record('forward0_quant'):
+        pass
+    with t.record('forward0_central_aggregation'):
+        pass
+    with t.record('grad_reduce'):
+        pass
+    row = t.epoch_rollup()
+    assert len(row) == 6
+    assert all(v >= 0 for v in row)
+    assert t.records == {}          # cleared per epoch
+
+
+def test_recorder_best(tmp_path):
+    r = Recorder()
+    r.add({'train': 0.5, 'val': 0.6, 'test': 0.55})
+    r.add({'train': 0.7, 'val': 0.8, 'test': 0.75})
+    r.add({'train': 0.9, 'val': 0.7, 'test': 0.95})
+    b = r.best()
+    assert b['epoch'] == 1 and b['val'] == 0.8 and b['test'] == 0.75
+    r.save(str(tmp_path), 'x')
+    assert (tmp_path / 'x_metrics.txt').exists()
+    assert (tmp_path / 'x_val_curve.pt').exists()
