@@ -65,7 +65,8 @@ def test_recorder_best(tmp_path):
     r.add({'train': 0.7, 'val': 0.8, 'test': 0.75})
     r.add({'train': 0.9, 'val': 0.7, 'test': 0.95})
     b = r.best()
-    assert b['epoch'] == 1 and b['val'] == 0.8 and b['test'] == 0.75
+    assert b['epoch'] == 1
+    assert abs(b['val'] - 0.8) < 1e-6 and abs(b['test'] - 0.75) < 1e-6
     r.save(str(tmp_path), 'x')
     assert (tmp_path / 'x_metrics.txt').exists()
     assert (tmp_path / 'x_val_curve.pt').exists()
