@@ -117,30 +117,71 @@ class SpmmView:
     critical path. Rows are split into <=SEG_EDGES-edge segments; rows
     with >1 segment combine via global atomics into pre-zeroed output
     rows (SURVEY.md §7 'hard parts': row-binning / load imbalance).
+
+    ``col_block`` > 0 additionally PERMUTES the edge array into
+    column-block-major order (segments keyed by (col/col_block, row)):
+    the grid sweeps one x column block at a time, so the block stays
+    LLC-resident while every consumer reads it — same kernel, different
+    work-item layout. Costs atomic combines for every row spanning >1
+    block.
     """
 
-    def __init__(self, indptr: Tensor, indices: Tensor, base: int, nrows: int):
+    def __init__(self, indptr: Tensor, indices: Tensor, base: int, nrows: int,
+                 col_block: int = 0):
         self.indptr = indptr
-        self.indices = indices
         self.base = int(base)
         self.nrows = int(nrows)
-        counts = (indptr[1:] - indptr[:-1]).cpu()
-        nseg = torch.clamp((counts + SEG_EDGES - 1) // SEG_EDGES, min=1)
-        seg_row = torch.repeat_interleave(torch.arange(nrows, dtype=torch.int64), nseg)
-        first = torch.repeat_interleave(torch.cumsum(nseg, 0) - nseg, nseg)
-        seg_in_row = torch.arange(seg_row.numel(), dtype=torch.int64) - first
-        e0 = indptr.cpu()[seg_row] + seg_in_row * SEG_EDGES
-        e1 = torch.minimum(e0 + SEG_EDGES, indptr.cpu()[seg_row + 1])
-        multi_mask = nseg > 1
-        self.seg_row = seg_row.to(torch.int32)
-        self.seg_e0 = e0
-        self.seg_e1 = e1
-        self.seg_multi = multi_mask[seg_row].to(torch.uint8)
-        self.zero_rows = torch.nonzero(multi_mask, as_tuple=True)[0].to(torch.int32)
+        indptr_c = indptr.cpu()
+        counts = (indptr_c[1:] - indptr_c[:-1])
+        if col_block > 0 and indices.numel():
+            idx_c = indices.cpu()
+            rows_pe = torch.repeat_interleave(
+                torch.arange(nrows, dtype=torch.int64), counts)
+            cb = idx_c // col_block
+            key = cb * nrows + rows_pe
+            order = torch.argsort(key, stable=True)
+            self.indices = idx_c[order].to(indices.device)
+            key = key[order]
+            # run starts: key change points, then chunk runs to SEG_EDGES
+            change = torch.ones(key.numel(), dtype=torch.bool)
+            change[1:] = key[1:] != key[:-1]
+            run_start = torch.nonzero(change, as_tuple=True)[0]
+            run_end = torch.cat([run_start[1:],
+                                 torch.tensor([key.numel()])])
+            run_len = run_end - run_start
+            nseg = (run_len + SEG_EDGES - 1) // SEG_EDGES
+            seg_runs = torch.repeat_interleave(
+                torch.arange(run_start.numel()), nseg)
+            first = torch.repeat_interleave(torch.cumsum(nseg, 0) - nseg, nseg)
+            in_run = torch.arange(seg_runs.numel(), dtype=torch.int64) - first
+            e0 = run_start[seg_runs] + in_run * SEG_EDGES
+            e1 = torch.minimum(e0 + SEG_EDGES, run_end[seg_runs])
+            seg_row = (key[run_start[seg_runs]] % nrows)
+            segs_per_row = torch.bincount(seg_row, minlength=nrows)
+            multi_mask = segs_per_row != 1     # incl. empty rows -> zeroed
+            self.seg_row = seg_row.to(torch.int32)
+            self.seg_e0 = e0
+            self.seg_e1 = e1
+            self.seg_multi = multi_mask[seg_row].to(torch.uint8)
+            self.zero_rows = torch.nonzero(multi_mask, as_tuple=True)[0].to(torch.int32)
+        else:
+            self.indices = indices
+            nseg = torch.clamp((counts + SEG_EDGES - 1) // SEG_EDGES, min=1)
+            seg_row = torch.repeat_interleave(torch.arange(nrows, dtype=torch.int64), nseg)
+            first = torch.repeat_interleave(torch.cumsum(nseg, 0) - nseg, nseg)
+            seg_in_row = torch.arange(seg_row.numel(), dtype=torch.int64) - first
+            e0 = indptr_c[seg_row] + seg_in_row * SEG_EDGES
+            e1 = torch.minimum(e0 + SEG_EDGES, indptr_c[seg_row + 1])
+            multi_mask = nseg > 1
+            self.seg_row = seg_row.to(torch.int32)
+            self.seg_e0 = e0
+            self.seg_e1 = e1
+            self.seg_multi = multi_mask[seg_row].to(torch.uint8)
+            self.zero_rows = torch.nonzero(multi_mask, as_tuple=True)[0].to(torch.int32)
         self._on = None
 
     def to(self, device):
-        for n in ('seg_row', 'seg_e0', 'seg_e1', 'seg_multi', 'zero_rows'):
+        for n in ('indices', 'seg_row', 'seg_e0', 'seg_e1', 'seg_multi', 'zero_rows'):
             setattr(self, n, getattr(self, n).to(device))
         return self
 

@@ -71,12 +71,16 @@ class GraphEngine:
         self.sage1_src_b = (din + 1.0).pow(-1.0)
 
         # decomposition views (zero-copy row-range splits + segmentation)
+        import os as _os
         from ..ops.kernels import SpmmView
-        self.full_view = SpmmView(g.indptr, g.indices, 0, g.num_inner)
+        cb = int(_os.environ.get('ADAQP_SPMM_COLBLOCK', '0'))
+        self.full_view = SpmmView(g.indptr, g.indices, 0, g.num_inner,
+                                  col_block=cb)
         cptr, cidx, _ = g.central_view()
-        self.central_view = SpmmView(cptr, cidx, 0, g.num_central)
+        self.central_view = SpmmView(cptr, cidx, 0, g.num_central, col_block=cb)
         mptr, midx, mbase = g.marginal_view()
-        self.marginal_view = SpmmView(mptr, midx, mbase, g.num_marginal)
+        self.marginal_view = SpmmView(mptr, midx, mbase, g.num_marginal,
+                                      col_block=cb)
         if self.device.type == 'cuda':
             for v in (self.full_view, self.central_view, self.marginal_view):
                 v.to(self.device)

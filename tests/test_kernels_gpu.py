@@ -296,3 +296,25 @@ def test_fused_dual_gemm_bf16():
     ref = a1.float() @ w1.float()
     assert torch.allclose(out.float(), ref, atol=2.0), \
         (out.float() - ref).abs().max()
+
+
+def test_spmm_blocked_matches_plain():
+    """Column-blocked edge ordering must give identical results (up to
+    fp reordering) to the plain row-major layout."""
+    from adaqp_amd.graph import random_partitioned_graph, partition_all
+    from adaqp_amd.ops.kernels import SpmmView, spmm
+    _native()
+    torch.manual_seed(13)
+    g = random_partitioned_graph(2000, 40000, 4, 3, 1, seed=31)
+    lg = partition_all(g, 1)[0]
+    F = 128
+    x = torch.randn(lg.num_nodes, F, device='cuda')
+    src = torch.rand(lg.num_nodes, device='cuda') + 0.5
+    dst = torch.rand(lg.num_inner, device='cuda') + 0.5
+    v0 = SpmmView(lg.indptr.cuda(), lg.indices.cuda(), 0, lg.num_inner).to('cuda')
+    v1 = SpmmView(lg.indptr.cuda(), lg.indices.cuda(), 0, lg.num_inner,
+                  col_block=256).to('cuda')
+    y0 = spmm(v0, x, None, src, dst)
+    y1 = spmm(v1, x, None, src, dst)
+    assert torch.allclose(y0, y1, atol=1e-3, rtol=1e-4), \
+        (y0 - y1).abs().max()

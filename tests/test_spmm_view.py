@@ -70,3 +70,32 @@ def test_recorder_best(tmp_path):
     r.save(str(tmp_path), 'x')
     assert (tmp_path / 'x_metrics.txt').exists()
     assert (tmp_path / 'x_val_curve.pt').exists()
+
+
+def test_blocked_view_covers_all_edges():
+    """col_block mode permutes edges; coverage must stay exact and the
+    CPU reference result (via edge expansion) must be preserved."""
+    for seed in range(2):
+        rows = 150
+        indptr, indices = _random_csr(rows, 40, seed, hub=2 * SEG_EDGES + 5)
+        v = SpmmView(indptr, indices, 0, rows, col_block=16)
+        # same multiset of (row, col) pairs
+        orig = []
+        for r in range(rows):
+            for e in range(int(indptr[r]), int(indptr[r + 1])):
+                orig.append((r, int(indices[e])))
+        perm = []
+        for i in range(v.seg_row.numel()):
+            r = int(v.seg_row[i])
+            for e in range(int(v.seg_e0[i]), int(v.seg_e1[i])):
+                perm.append((r, int(v.indices[e])))
+        assert sorted(orig) == sorted(perm)
+        # single-segment rows store plainly; all others zeroed+atomic
+        from collections import Counter
+        per_row = Counter(int(r) for r in v.seg_row)
+        zr = set(v.zero_rows.tolist())
+        for r in range(rows):
+            if per_row.get(r, 0) == 1:
+                assert r not in zr
+            else:
+                assert r in zr

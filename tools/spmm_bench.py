@@ -34,13 +34,15 @@ def main():
     p.add_argument('--deg', type=int, default=50)
     p.add_argument('--F', type=int, default=256)
     p.add_argument('--iters', type=int, default=10)
+    p.add_argument('--col-block', type=int, default=0)
     args = p.parse_args()
     assert torch.cuda.is_available()
     from adaqp_amd.ops.kernels import SpmmView, spmm, native
     native()
     dev = torch.device('cuda')
     indptr, indices, E = build_csr(args.rows, args.deg, args.cols, dev)
-    view = SpmmView(indptr, indices, 0, args.rows).to(dev)
+    view = SpmmView(indptr, indices, 0, args.rows,
+                    col_block=args.col_block).to(dev)
     x = torch.randn(args.cols, args.F, device=dev)
     src = torch.rand(args.cols, device=dev) + 0.5
     dst = torch.rand(args.rows, device=dev) + 0.5
@@ -53,7 +55,7 @@ def main():
     torch.cuda.synchronize()
     dt = (time.perf_counter() - t0) / args.iters
     gather_bytes = E * (args.F * 4 + 12)
-    print(f'rows={args.rows} E={E} F={args.F}: {dt*1e3:.3f} ms/call, '
+    print(f'rows={args.rows} E={E} F={args.F} cb={args.col_block}: {dt*1e3:.3f} ms/call, '
           f'apparent gather {gather_bytes/dt/1e12:.2f} TB/s')
     sys.stdout.flush()
 
