@@ -13,7 +13,7 @@ pytestmark = pytest.mark.gpu
 P = 2
 
 
-def _worker(rank, world, port, mode, q):
+def _worker(rank, world, port, mode, q, dtype='fp32'):
     os.environ.update(MASTER_ADDR='127.0.0.1', MASTER_PORT=str(port),
                       RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK='0')
     import torch
@@ -36,6 +36,8 @@ def _worker(rank, world, port, mode, q):
         lg = partition_all(g, world)[rank]
         engine = GraphEngine(lg, RunMode(mode), DistGNNType.DistGCN,
                              msg_dims=[32, 32, 32], device=dev)
+        if dtype == 'bf16':
+            engine.compute_dtype = torch.bfloat16
         if engine.bit_type.name == 'QUANT':
             engine.set_uniform_assignment(8)
         torch.manual_seed(33)
@@ -51,12 +53,16 @@ def _worker(rank, world, port, mode, q):
         Communicator.shutdown()
 
 
-@pytest.mark.parametrize('mode', ['Vanilla', 'AdaQP', 'AdaQP-q'])
-def test_two_ranks_one_gpu(mode):
+CASES = [('Vanilla', 'fp32'), ('AdaQP', 'fp32'), ('AdaQP-q', 'fp32'),
+         ('AdaQP', 'bf16')]
+
+
+@pytest.mark.parametrize('mode,dtype', CASES)
+def test_two_ranks_one_gpu(mode, dtype):
     ctx = mp.get_context('spawn')
     q = ctx.SimpleQueue()
-    port = 29540 + ['Vanilla', 'AdaQP', 'AdaQP-q'].index(mode)
-    procs = [ctx.Process(target=_worker, args=(r, P, port, mode, q))
+    port = 29540 + CASES.index((mode, dtype))
+    procs = [ctx.Process(target=_worker, args=(r, P, port, mode, q, dtype))
              for r in range(P)]
     for p in procs:
         p.start()
