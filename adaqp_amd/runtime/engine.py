@@ -49,7 +49,8 @@ class GraphEngine:
         self.base_seed = base_seed
         self.compute_dtype = torch.float32   # set to torch.bfloat16 for bf16 mode
         self._rng_counter = 0
-        self.timer = Timer(enabled=False, cuda=self.device.type == 'cuda')
+        self.timer = Timer(enabled=False, cuda=self.device.type == 'cuda',
+                           mode='events')
         self.is_tracing = False
         self.traced: Dict[str, Tensor] = {}      # key -> accumulated per-send-node variance proxy
 

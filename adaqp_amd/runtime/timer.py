@@ -1,31 +1,47 @@
 """Span timer with the reference's taxonomy (``AdaQP/util/timer.py``).
 
-Two backends: wall-clock with stream-sync fences (reference-style, for
-per-epoch breakdowns) and disabled (default — fences serialize the
-overlap the framework exists to create, so production runs keep it off
-and use rocprofv3 for kernel-level evidence).
+Three backends:
+- disabled (default): production runs keep timing off and use rocprofv3.
+- ``mode='sync'``: wall clock with stream-sync fences (reference-style,
+  ``timer.py:18-27``) — accurate totals but SERIALIZES the comm/compute
+  overlap it measures.
+- ``mode='events'``: hipEvent pairs recorded on the current stream; spans
+  are resolved at epoch rollup with one sync — measures GPU span time
+  without perturbing the overlap (the MI355X-native replacement the
+  survey calls for).
 """
 from __future__ import annotations
 
 import time
 from collections import defaultdict
 from contextlib import contextmanager
-from typing import Dict, List
+from typing import Dict, List, Tuple
 
 import torch
 
 
 class Timer:
-    def __init__(self, enabled: bool = False, cuda: bool = False):
+    def __init__(self, enabled: bool = False, cuda: bool = False,
+                 mode: str = 'sync'):
         self.enabled = enabled
         self.cuda = cuda
+        self.mode = mode if cuda else 'sync'
         self.records: Dict[str, float] = {}
+        self.pending: List[Tuple[str, torch.cuda.Event, torch.cuda.Event]] = []
         self.epoch_rows: List[List[float]] = []
 
     @contextmanager
     def record(self, name: str):
         if not self.enabled:
             yield
+            return
+        if self.cuda and self.mode == 'events':
+            e0 = torch.cuda.Event(enable_timing=True)
+            e1 = torch.cuda.Event(enable_timing=True)
+            e0.record()
+            yield
+            e1.record()
+            self.pending.append((name, e0, e1))
             return
         if self.cuda:
             torch.cuda.synchronize()
@@ -35,12 +51,24 @@ class Timer:
             torch.cuda.synchronize()
         self.records[name] = self.records.get(name, 0.0) + time.perf_counter() - t0
 
+    def _drain_events(self):
+        if not self.pending:
+            return
+        torch.cuda.synchronize()
+        for name, e0, e1 in self.pending:
+            self.records[name] = (self.records.get(name, 0.0)
+                                  + e0.elapsed_time(e1) / 1e3)
+        self.pending = []
+
     def clear(self):
         self.records = {}
+        self.pending = []
 
     def epoch_rollup(self) -> List[float]:
-        """[total_comm, quant+dequant, central_agg, marginal_agg, full_agg]
-        (reference ``timer.py:29-51``)."""
+        """[comm, quant+dequant, central_agg, marginal_agg, full_agg,
+        grad_reduce] (reference ``timer.py:29-51``)."""
+        if self.cuda and self.mode == 'events':
+            self._drain_events()
         buckets = defaultdict(float)
         for k, v in self.records.items():
             if 'exchange' in k:

@@ -177,9 +177,12 @@ def test_e2e_train_step_gpu():
         model = DistGCN(32, 16, 7, 3).to(comm.device)
         opt = torch.optim.Adam(model.parameters())
         gc = global_train_count(engine)
+        engine.timer.enabled = True    # event-based spans must not break
         l0 = float(train_epoch(engine, model, opt, gc, False))
         for _ in range(20):
             l = float(train_epoch(engine, model, opt, gc, False))
+        row = engine.timer.epoch_rollup()
+        assert sum(row) > 0, 'event timer recorded nothing'
         assert torch.isfinite(torch.tensor(l))
         assert l < l0  # training must reduce loss on a learnable graph
     finally:
