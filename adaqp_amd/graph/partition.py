@@ -97,6 +97,15 @@ def build_local_graph(g: GlobalGraph, assign: Tensor, rank: int,
                       out_deg: Optional[Tensor] = None) -> LocalGraph:
     if in_deg is None or out_deg is None:
         in_deg, out_deg = global_degrees(g)
+    if not torch.equal(in_deg, out_deg):
+        # backward reuses the forward CSR with swapped norms, which is the
+        # exact transpose ONLY for bidirected graphs (every reference
+        # dataset is symmetrized + self-looped; so is the synthetic
+        # generator). Degree equality is the cheap necessary condition.
+        raise ValueError(
+            'graph is not bidirected (in/out degree mismatch); symmetrize '
+            'it first (add both edge directions) — required for exact '
+            'backward aggregation')
 
     # my in-edges: dst owned by me
     emask = assign[g.dst] == rank

@@ -45,6 +45,11 @@ def parse_args():
     p.add_argument('--layers', type=int, default=3)
     p.add_argument('--assign-bits', type=int, default=4,
                    help='uniform bit width for quantized modes')
+    p.add_argument('--assign-scheme', type=str, default='uniform',
+                   choices=['uniform', 'adaptive'],
+                   help='adaptive runs the cost profiler + HiGHS MILP at '
+                        'setup (reference default); uniform keeps short '
+                        'benches deterministic')
     p.add_argument('--part-dir', type=str, default='part_data_bench')
     p.add_argument('--cpu', action='store_true', help='force CPU (debug)')
     p.add_argument('--dtype', type=str, default='fp32', choices=['fp32', 'bf16'],
@@ -115,7 +120,14 @@ def main():
     if args.dtype == 'bf16':
         engine.compute_dtype = torch.bfloat16
     if mode.bit_type.name == 'QUANT':
-        engine.set_uniform_assignment(args.assign_bits)
+        if args.assign_scheme == 'adaptive':
+            from adaqp_amd.assigner import Assigner
+            from adaqp_amd.helpers import AssignScheme
+            assigner = Assigner(engine, AssignScheme.ADAPTIVE,
+                                init_bits=args.assign_bits)
+            assigner.initial_assignment()
+        else:
+            engine.set_uniform_assignment(args.assign_bits)
 
     torch.manual_seed(12345)
     if args.model == 'gcn':
