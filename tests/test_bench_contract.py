@@ -46,3 +46,9 @@ def test_bench_json_contract_1rank_defaults(tmp_path):
     assert out.returncode == 0, out.stderr[-2000:]
     d = json.loads([l for l in out.stdout.splitlines() if l.startswith('{')][0])
     assert d['n_gpus'] == 1
+
+
+def test_bench_adaptive_scheme_2rank(tmp_path):
+    d = _run_bench(2, ['--part-dir', str(tmp_path / 'parts'),
+                       '--assign-scheme', 'adaptive'])
+    assert d['config']['mode'] == 'AdaQP'
