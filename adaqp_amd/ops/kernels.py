@@ -131,6 +131,7 @@ class SpmmView:
         self.indptr = indptr
         self.base = int(base)
         self.nrows = int(nrows)
+        self.col_blocked = col_block > 0 and indices.numel() > 0
         indptr_c = indptr.cpu()
         counts = (indptr_c[1:] - indptr_c[:-1])
         if col_block > 0 and indices.numel():
@@ -195,6 +196,10 @@ def spmm(view: 'SpmmView', x_local: Tensor, x_remote: Optional[Tensor],
     src_scale: [n_local+n_remote] or None; dst_scale: [nrows] or None.
     """
     indptr, indices, num_rows = view.indptr, view.indices, view.nrows
+    if not x_local.is_cuda and view.col_blocked:
+        raise RuntimeError('column-blocked SpmmView is GPU-only (the CPU '
+                           'fallback consumes indptr, which the blocked '
+                           'layout does not preserve)')
     if x_local.is_cuda:
         out = torch.empty(num_rows, x_local.shape[1], dtype=x_local.dtype,
                           device=x_local.device)

@@ -74,7 +74,8 @@ class GraphEngine:
         # decomposition views (zero-copy row-range splits + segmentation)
         import os as _os
         from ..ops.kernels import SpmmView
-        cb = int(_os.environ.get('ADAQP_SPMM_COLBLOCK', '0'))
+        cb = (int(_os.environ.get('ADAQP_SPMM_COLBLOCK', '0'))
+              if self.device.type == 'cuda' else 0)
         self.full_view = SpmmView(g.indptr, g.indices, 0, g.num_inner,
                                   col_block=cb)
         cptr, cidx, _ = g.central_view()
