@@ -36,6 +36,11 @@ def fit_cost_models(comm: Communicator, num_points: int = 12,
     if W == 1:
         return models
     dev = comm.device
+    # gloo cannot carry CUDA tensors over p2p; in the gloo-staged debug
+    # transport the real exchange stages via host memory anyway, so the
+    # cost model should measure host-side gloo too.
+    if dev.type == 'cuda' and 'nccl' not in comm.backend:
+        dev = torch.device('cpu')
     sizes = _sizes(num_points)
     buf = torch.empty(sizes[-1], dtype=torch.uint8, device=dev)
     mins: Dict[Tuple[int, int], List[float]] = {}
