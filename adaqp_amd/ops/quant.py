@@ -64,6 +64,8 @@ def qparams(x: Tensor, bits: int) -> Tuple[Tensor, Tensor]:
 
 
 def bytes_per_node(num_feats: int, bits: int) -> int:
+    """Payload bytes per node; must agree with comm.buffers and the HIP
+    kernels' per-node padding (buffer.py:181-186 analogue)."""
     return (num_feats * bits + 7) // 8
 
 
