@@ -4,7 +4,7 @@ Reference parity: ``quant_cuda.pack/unpack_single_precision``
 (``/root/reference/AdaQP/util/quantization/src/quantization_cuda_kernel.cu:34-156``)
 plus the Python min/max + scale glue (``AdaQP/model/op_util.py:20-67``).
 
-MI355X redesign (the HIP kernel in ``csrc/quant.hip`` implements the
+MI355X redesign (the HIP kernels in ``csrc/kernels.hip`` implements the
 same math bit-for-bit; this torch version is the CPU path AND the test
 oracle):
 
