@@ -53,9 +53,15 @@ class GlobalGraph:
 
 
 def _dedup_edges(src: Tensor, dst: Tensor, n: int) -> Tuple[Tensor, Tensor]:
-    key = src * n + dst
+    """Deduplicate (src,dst) pairs. The unique/sort over E*2 int64 keys is
+    the generator's hot spot (amazonProducts: ~5*10^8 keys, minutes on
+    CPU) — run it on the GPU when one is present (seconds; the result is
+    deterministic either way)."""
+    dev = 'cuda' if torch.cuda.is_available() else 'cpu'
+    key = (src.to(dev) * n + dst.to(dev)).contiguous()
     key = torch.unique(key)
-    return key // n, key % n
+    out = key // n, key % n
+    return out[0].cpu(), out[1].cpu()
 
 
 def synth_graph(name: str, num_parts: int, seed: int = 0, cut_frac: float = 0.10,
