@@ -176,14 +176,16 @@ __global__ void quant_unpack_kernel(
 // ---------------------------------------------------------------------------
 // spmm_csr: y[r] = dst_scale[r] * sum_{e in row r} src_scale[c_e] * x[c_e]
 //
-// One SUB-wavefront (SW lanes, 16/32/64 picked from F) per destination row —
-// a 64-wide wave covers 64/SW rows so narrow feature dims (ogbn-products
-// F=100 -> SW=32) keep every lane busy. Lane owns features [4*sl, 4*sl+4)
-// (float4) per chunk of 4*SW. Two edges are accumulated in flight (dual
-// accumulators) to cover gather latency. Dual-tensor input: columns
-// >= n_local read the REMOTE block directly (no [N,F] concat per layer).
-// XCD-aware bijective block swizzle gives each XCD a contiguous row chunk
-// so clustered neighbor rows hit the same L2 (guide §5.5 T1).
+// One SUB-wavefront (SW in {16,32,64}, chosen to cover F) per destination-row
+// SEGMENT (seg_* arrays; hub rows are split to <=SEG_EDGES edges and combined
+// with atomics into pre-zeroed rows). Lane sl owns features
+// [VE*sl, VE*sl+VE) per chunk of VE*SW, VE picked host-side so the
+// VE*sizeof(T)-byte row-base loads stay aligned (VE | F). Two edges are
+// accumulated in flight (dual accumulators) to cover gather latency.
+// Dual-tensor input: columns >= n_local read the REMOTE block directly
+// (no [N,F] concat per layer). XCD-aware bijective block swizzle gives
+// each XCD a contiguous chunk of work items so clustered neighbor rows
+// hit the same L2 (guide §5.5 T1).
 // ---------------------------------------------------------------------------
 template <typename T>
 __global__ void zero_rows_kernel(T* __restrict__ y,
