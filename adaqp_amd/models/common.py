@@ -117,8 +117,10 @@ def fused_dual_linear_ok(x: Tensor, h: Tensor, n_out: int) -> bool:
     if os.environ.get('ADAQP_FUSED_SAGE') != '1':
         return False
     from ..ops.kernels import has_native
-    return (x.is_cuda and x.dtype == torch.bfloat16
-            and h.dtype == torch.bfloat16
+    # fused_dual_linear casts both operands to bf16, so fp32 inputs are
+    # acceptable (layer-0 features stay fp32 under a bf16 engine).
+    ok_dt = (torch.bfloat16, torch.float32)
+    return (x.is_cuda and x.dtype in ok_dt and h.dtype in ok_dt
             and n_out % 16 == 0 and n_out <= 256
             and x.shape[1] % 8 == 0 and h.shape[1] % 8 == 0
             and has_native())

@@ -34,9 +34,8 @@ class DistSAGEConv(nn.Module):
         if self.aggregator_type == 'mean':
             x_self = x[:engine.graph.num_inner]
             if (engine.compute_dtype == torch.bfloat16
-                    and fused_dual_linear_ok(h_neigh, h_neigh,
-                                             self.fc_neigh.weight.shape[1])
-                    and x_self.shape[1] % 8 == 0):
+                    and fused_dual_linear_ok(x_self, h_neigh,
+                                             self.fc_neigh.weight.shape[1])):
                 return fused_dual_linear(x_self, h_neigh,
                                          self.fc_self, self.fc_neigh)
             return self.fc_self(x_self) + self.fc_neigh(h_neigh)

@@ -15,7 +15,12 @@ oracle):
   (coalesced 128B stores on CDNA4), vs the reference's node-axis packing.
   Each node occupies ceil(F*bits/8) bytes.
 - RNG: stateless counter hash of (seed, node_tag, feature) — no curand
-  state; identical on CPU/HIP so packings agree bit-for-bit.
+  state; the SAME noise sequence on CPU/HIP, so packings are statistically
+  identical and byte-identical except where FP contraction (FMA) on the
+  GPU lands a value exactly on a rounding boundary (< 0.5% of bytes in
+  tests/test_kernels_gpu.py). Harmless: only the sender's kernel ever
+  produces a given payload, so there is no cross-device mismatch on the
+  wire.
 - scale is rounded to bf16 BEFORE quantizing, so the receiver's bf16
   dequant is the exact inverse (the reference quantizes with the fp32
   scale but ships bf16 — a small bias it tolerates).

@@ -9,9 +9,14 @@ import torch
 
 class Recorder:
     def __init__(self):
-        self.rows: List[List[float]] = []     # [train, val, test] per epoch
+        self.rows: List[List[float]] = []     # [train, val, test] per eval
+        self.epochs: List[int] = []           # training epoch of each eval
 
-    def add(self, metrics: Dict[str, float]):
+    def add(self, metrics: Dict[str, float], epoch: Optional[int] = None):
+        """Record one evaluation. ``epoch`` is the TRAINING epoch it was
+        computed at (evals may be sparse with eval_every > 1); defaults to
+        the row index for dense evaluation."""
+        self.epochs.append(len(self.rows) if epoch is None else int(epoch))
         self.rows.append([metrics['train'], metrics['val'], metrics['test']])
 
     def best(self) -> Dict[str, float]:
@@ -19,7 +24,8 @@ class Recorder:
             return {'epoch': -1, 'val': 0.0, 'test': 0.0}
         t = torch.tensor(self.rows)
         i = int(t[:, 1].argmax())
-        return {'epoch': i, 'val': float(t[i, 1]), 'test': float(t[i, 2])}
+        return {'epoch': self.epochs[i], 'val': float(t[i, 1]),
+                'test': float(t[i, 2])}
 
     def save(self, out_dir: str, tag: str, extra: Optional[Dict] = None):
         os.makedirs(out_dir, exist_ok=True)

@@ -66,7 +66,18 @@ class Timer:
 
     def epoch_rollup(self) -> List[float]:
         """[comm, quant+dequant, central_agg, marginal_agg, full_agg,
-        grad_reduce] (reference ``timer.py:29-51``)."""
+        grad_reduce] (reference ``timer.py:29-51``).
+
+        SEMANTICS (events mode): each value is the summed GPU *span* time
+        of that bucket, measured per-stream without serializing. Spans on
+        different streams can OVERLAP in wall time — by design, the comm
+        bucket (comm stream) runs concurrently with central_agg (default
+        stream) in the decomposed path — so the row may sum to MORE than
+        the epoch's wall time. Read it as span accounting, not as a
+        partition of the epoch: wall saved by overlap shows up as
+        (sum of buckets) − (epoch wall time). In 'sync' mode fences
+        serialize every span, so the row does partition wall time but the
+        overlap being measured is destroyed (reference behavior)."""
         if self.cuda and self.mode == 'events':
             self._drain_events()
         buckets = defaultdict(float)

@@ -9,6 +9,7 @@ Communicator -> GraphEngine -> Assigner -> initial assignment -> Model.
 """
 from __future__ import annotations
 
+import json
 import logging
 import os
 import time
@@ -72,8 +73,12 @@ class Trainer:
         fmt = f'[rank {self.comm.rank}] %(asctime)s %(levelname)s %(message)s'
         logging.basicConfig(level=getattr(logging, level.upper(), logging.INFO),
                             format=fmt)
-        # file log, reference parity (runtime_util.py:22-32)
-        fh = logging.FileHandler('trainer.log')
+        # file log, reference parity (runtime_util.py:22-32) — written under
+        # the run's exp/ output tree, not the CWD, so runs don't dirty it
+        log_dir = os.path.join(getattr(self.args, 'exp_dir', None) or 'exp',
+                               'logs')
+        os.makedirs(log_dir, exist_ok=True)
+        fh = logging.FileHandler(os.path.join(log_dir, 'trainer.log'))
         fh.setFormatter(logging.Formatter(fmt))
         logging.getLogger('trainer').addHandler(fh)
 
@@ -83,17 +88,33 @@ class Trainer:
         part_dir = getattr(args, 'partition_dir', None) or 'part_data'
         ds = args.dataset
         meta = os.path.join(part_dir, ds, f'{world}part', f'{ds}.json')
+        dcfg = self.cfg.get('data', {})
+        # only a CLI-provided --scale is an explicit request; the config's
+        # data.scale is a generation default and never invalidates a cache
+        req_scale = getattr(args, 'scale', None)
         if not os.path.exists(meta):
+            scale = req_scale if req_scale is not None else dcfg.get('scale', 1.0)
             if rank == 0:
                 logger.info('partition cache miss -> generating synthetic '
                             f'{ds} and partitioning into {world}')
-                dcfg = self.cfg.get('data', {})
-                scale = getattr(args, 'scale', None) or dcfg.get('scale', 1.0)
                 g = synth_graph(ds, world, seed=17,
                                 cut_frac=dcfg.get('cut_frac', 0.1),
                                 scale=scale)
-                save_partitions(partition_all(g, world), part_dir, ds)
+                save_partitions(partition_all(g, world), part_dir, ds,
+                                meta={'scale': scale})
             self.comm.barrier()
+        elif req_scale is not None:
+            # a cached partition generated at a different --scale must not
+            # be silently reused: validate the meta record.
+            # With no explicit scale requested, the cached one is adopted.
+            with open(meta) as f:
+                cached_scale = json.load(f).get('scale', 1.0)
+            if cached_scale != req_scale:
+                raise SystemExit(
+                    f'cached partition under {os.path.dirname(meta)} was '
+                    f'built at scale={cached_scale}, but this run requests '
+                    f'scale={req_scale}; delete the cache dir or pass a '
+                    f'different --partition_dir')
         try:
             self.graph = load_partition(part_dir, ds, world, rank)
         except FileNotFoundError as e:
@@ -163,14 +184,17 @@ class Trainer:
             self.epoch_times.append(time.perf_counter() - t0)
             if self.engine.timer.enabled:
                 self.engine.timer.epoch_rollup()
-            if epoch % rt.get('eval_every', 1) == 0:
+            evaluated = epoch % rt.get('eval_every', 1) == 0
+            if evaluated:
                 metrics = evaluate(self.engine, self.model, self.multilabel)
-                self.recorder.add(metrics)
+                self.recorder.add(metrics, epoch)
             if epoch % rt['log_steps'] == 0 and self.comm.rank == 0:
+                # metrics are only printed for epochs they were computed at
+                # (with eval_every > 1 they would otherwise be stale)
+                acc = (f'train {metrics["train"]:.4f} val {metrics["val"]:.4f} '
+                       f'test {metrics["test"]:.4f} ' if evaluated else '')
                 logger.info(
-                    f'epoch {epoch:04d} loss {float(loss):.4f} '
-                    f'train {metrics["train"]:.4f} val {metrics["val"]:.4f} '
-                    f'test {metrics["test"]:.4f} '
+                    f'epoch {epoch:04d} loss {float(loss):.4f} {acc}'
                     f'epoch_time {self.epoch_times[-1]*1e3:.1f}ms')
             if ckpt_every and ckpt_path and (epoch + 1) % ckpt_every == 0:
                 self.save_checkpoint(ckpt_path, epoch + 1)

@@ -161,10 +161,13 @@ def main():
 
     if rank == 0:
         n, e, _, _, _ = DATASET_SHAPES[args.dataset]
+        # metric label is WORLD-SIZE-ACCURATE: "8-part" is printed only
+        # when 8 ranks actually ran (at world=1 there are no boundary
+        # nodes, so quant/exchange are no-ops — a 1-GPU number is a
+        # full-graph compute number, not an AdaQP N-part number).
+        mname = 'GraphSAGE' if args.model == 'sage' else 'GCN'
         out = {
-            'metric': 'per-epoch time (s), ogbn-products GraphSAGE 8-part'
-                      if args.dataset == 'ogbn-products' and args.model == 'sage'
-                      else f'per-epoch time (s), {args.dataset} {args.model} {world}-part',
+            'metric': f'per-epoch time (s), {args.dataset} {mname} {world}-part',
             'value': per_epoch,
             'unit': 's/epoch',
             'n_gpus': world,

@@ -32,6 +32,9 @@ def main():
     p.add_argument('--num_parts', type=int, default=None,
                    help='expected partition count; validated against WORLD_SIZE')
     p.add_argument('--partition_dir', type=str, default='part_data')
+    p.add_argument('--scale', type=float, default=None,
+                   help='synthetic graph scale; validated against the '
+                        'partition cache (omit to adopt the cached scale)')
     p.add_argument('--num_epochs', type=int, default=None)
     p.add_argument('--lr', type=float, default=None)
     p.add_argument('--log_steps', type=int, default=None)
