@@ -18,7 +18,12 @@ def main():
                    choices=['reddit', 'yelp', 'ogbn-products', 'amazonProducts'])
     p.add_argument('--partition_dir', type=str, default='part_data')
     p.add_argument('--partition_size', type=int, default=4)
-    p.add_argument('--method', type=str, default='range', choices=['range', 'bfs'])
+    p.add_argument('--method', type=str, default='range',
+                   choices=['range', 'grow', 'bfs'],
+                   help="'range' exploits the synthetic generator's planted "
+                        "locality (METIS-quality there, O(1)); 'grow' is the "
+                        "vectorized balanced region-growing partitioner for "
+                        "arbitrary graphs ('bfs' is its back-compat alias)")
     p.add_argument('--cut_frac', type=float, default=0.10)
     p.add_argument('--scale', type=float, default=1.0)
     p.add_argument('--seed', type=int, default=17)
