@@ -1,5 +1,6 @@
 from .csr import LocalGraph, coo_to_csr, pad_feat_dim
 from .synthetic import GlobalGraph, synth_graph, random_partitioned_graph, tiny_ring_graph, DATASET_SHAPES
+from .ingest import load_graph_dir
 from .partition import (range_assignment, bfs_assignment, grow_assignment,
                         edge_cut, build_local_graph,
                         partition_all, save_partitions, load_partition, global_degrees)
@@ -9,5 +10,5 @@ __all__ = [
     'random_partitioned_graph', 'tiny_ring_graph', 'DATASET_SHAPES',
     'range_assignment', 'bfs_assignment', 'grow_assignment', 'edge_cut',
     'build_local_graph', 'partition_all',
-    'save_partitions', 'load_partition', 'global_degrees',
+    'save_partitions', 'load_partition', 'global_degrees', 'load_graph_dir',
 ]

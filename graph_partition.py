@@ -16,6 +16,11 @@ def main():
     p = argparse.ArgumentParser()
     p.add_argument('--dataset', type=str, default='reddit',
                    choices=['reddit', 'yelp', 'ogbn-products', 'amazonProducts'])
+    p.add_argument('--raw_dir', type=str, default=None,
+                   help='load a REAL dataset from this GraphSAINT-layout '
+                        'directory (adj_full.npz + feats.npy + class_map/'
+                        'labels + role.json) instead of generating a '
+                        'synthetic graph (reference --raw_dir parity)')
     p.add_argument('--partition_dir', type=str, default='part_data')
     p.add_argument('--partition_size', type=int, default=4)
     p.add_argument('--method', type=str, default='range',
@@ -29,8 +34,15 @@ def main():
     p.add_argument('--seed', type=int, default=17)
     args = p.parse_args()
 
-    g = synth_graph(args.dataset, args.partition_size, seed=args.seed,
-                    cut_frac=args.cut_frac, scale=args.scale)
+    if args.raw_dir:
+        from adaqp_amd.graph import load_graph_dir
+        g = load_graph_dir(args.raw_dir)
+        if args.method == 'range':
+            # contiguous ranges carry no locality on arbitrary data
+            args.method = 'grow'
+    else:
+        g = synth_graph(args.dataset, args.partition_size, seed=args.seed,
+                        cut_frac=args.cut_frac, scale=args.scale)
     parts = partition_all(g, args.partition_size, method=args.method)
     d = save_partitions(parts, args.partition_dir, args.dataset,
                         meta={'method': args.method, 'cut_frac': args.cut_frac,
