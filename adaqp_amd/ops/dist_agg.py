@@ -19,7 +19,6 @@ normalization performs the cross-partition reduction — exact transpose
 """
 from __future__ import annotations
 
-from contextlib import nullcontext
 from typing import Optional, Tuple
 
 import torch
@@ -34,40 +33,62 @@ from .kernels import mixed_quantize, mixed_dequantize, spmm
 # boundary exchange
 # --------------------------------------------------------------------------
 
-def fp_exchange(engine, x_local: Tensor, key: str) -> Tensor:
+def _exchange_start(engine, x_local: Tensor, key: str, quant: bool):
+    """Device-side producer work only (gather or quantize kernels) — no
+    transport. Split from :func:`_exchange_finish` so the decomposed path
+    can enqueue the central aggregation BETWEEN the two: the transport may
+    block the host (gloo staging), and anything enqueued after it would
+    lose the overlap."""
+    if quant:
+        plan = engine.plans[key]
+        seed = engine.next_seed()
+        if engine.is_tracing:
+            send = x_local.index_select(0, engine.graph.total_send_idx)
+            engine.trace(key, send)
+        with engine.timer.record(f'{key}_quant'):
+            payload, params = mixed_quantize(x_local, plan.send, seed)
+        return ('qt', payload, params, x_local.dtype)
+    send = x_local.index_select(0, engine.graph.total_send_idx)
+    if engine.is_tracing:
+        engine.trace(key, send)
+    return ('fp', send, None, x_local.dtype)
+
+
+def _exchange_finish(engine, staged, key: str) -> Tensor:
+    """Transport + consumer-side kernels (dequant/scatter)."""
     comm = Communicator.ctx
     g = engine.graph
-    send = x_local.index_select(0, g.total_send_idx)
-    if engine.is_tracing:
-        engine.trace(key, send)
-    with engine.timer.record(f'{key}_exchange'):
-        recv, _ = comm.exchange_rows(send, g.send_splits, g.recv_splits)
-    return recv
-
-
-def qt_exchange(engine, x_local: Tensor, key: str) -> Tensor:
-    comm = Communicator.ctx
+    kind, a, b, dtype = staged
+    if kind == 'fp':
+        with engine.timer.record(f'{key}_exchange'):
+            recv, _ = comm.exchange_rows(a, g.send_splits, g.recv_splits)
+        return recv
     plan = engine.plans[key]
-    seed = engine.next_seed()
-    if engine.is_tracing:
-        send = x_local.index_select(0, engine.graph.total_send_idx)
-        engine.trace(key, send)
-    with engine.timer.record(f'{key}_quant'):
-        payload, params = mixed_quantize(x_local, plan.send, seed)
+    payload, params = a, b
     recv_payload = torch.empty(plan.recv.total_bytes, dtype=torch.uint8,
-                               device=x_local.device)
+                               device=payload.device)
     recv_params = torch.empty(2 * plan.recv.total_nodes, dtype=torch.bfloat16,
-                              device=x_local.device)
+                              device=payload.device)
     with engine.timer.record(f'{key}_exchange'):
         comm.all_to_all_v(recv_payload, payload,
                           plan.recv.byte_splits, plan.send.byte_splits)
         comm.all_to_all_v(recv_params, params,
                           plan.recv.param_splits, plan.send.param_splits)
-    out = torch.empty(engine.graph.num_remote, plan.F, dtype=x_local.dtype,
-                      device=x_local.device)
+    out = torch.empty(g.num_remote, plan.F, dtype=dtype,
+                      device=payload.device)
     with engine.timer.record(f'{key}_dequant'):
         mixed_dequantize(recv_payload, recv_params, plan.recv, out)
     return out
+
+
+def fp_exchange(engine, x_local: Tensor, key: str) -> Tensor:
+    return _exchange_finish(engine,
+                            _exchange_start(engine, x_local, key, False), key)
+
+
+def qt_exchange(engine, x_local: Tensor, key: str) -> Tensor:
+    return _exchange_finish(engine,
+                            _exchange_start(engine, x_local, key, True), key)
 
 
 def _exchange(engine, x_local: Tensor, key: str, is_train: bool) -> Tensor:
@@ -129,32 +150,43 @@ def full_propagation(engine, x_local: Tensor, key: str, is_train: bool,
 def decomposed_propagation(engine, x_local: Tensor, key: str, is_train: bool,
                            mode: PropagationMode) -> Tensor:
     """Central aggregation (default stream) overlapped with
-    quant->all_to_all->dequant (comm stream)."""
+    quant->all_to_all->dequant (comm stream).
+
+    Host-call order matters: producer kernels (quantize/gather) are
+    enqueued on the comm stream, then the central SpMM is enqueued on the
+    default stream, and only THEN the transport runs. With RCCL the
+    transport is stream-ordered device work so any order would overlap;
+    with a host-blocking transport (gloo staging — several ranks sharing
+    one GPU, or CPU debug) the central SpMM must already be in the GPU
+    queue when the host blocks, otherwise the overlap the decomposition
+    exists for never happens (the reference needed a side thread for
+    this, ``ops.py:156-193``; stream order + one event replaces it)."""
     src_scale, dst_scale, add_self = _scales(engine, mode)
     g = engine.graph
     on_gpu = x_local.is_cuda
+    quant = engine.bit_type == BitType.QUANT and is_train
 
     if on_gpu:
         cur = torch.cuda.current_stream()
         engine.comm_stream.wait_stream(cur)          # x_local is ready
         x_local.record_stream(engine.comm_stream)
         with torch.cuda.stream(engine.comm_stream):
-            remote = _exchange(engine, x_local, key, is_train)
-            engine.remote_ready.record(engine.comm_stream)
-        ctx = nullcontext()
-    else:
-        remote = _exchange(engine, x_local, key, is_train)
-        ctx = nullcontext()
-
-    with ctx:
-        # central rows only touch local columns -> safe to run now
+            staged = _exchange_start(engine, x_local, key, quant)
+        # central rows only touch local columns -> enqueue NOW, before
+        # the transport can block the host
         with engine.timer.record(f'{key}_central_aggregation'):
             y_c = _agg(engine, engine.central_view, x_local, None, src_scale,
                        dst_scale)
-
-    if on_gpu:
+        with torch.cuda.stream(engine.comm_stream):
+            remote = _exchange_finish(engine, staged, key)
+            engine.remote_ready.record(engine.comm_stream)
         torch.cuda.current_stream().wait_event(engine.remote_ready)
         remote.record_stream(torch.cuda.current_stream())
+    else:
+        remote = _exchange(engine, x_local, key, is_train)
+        with engine.timer.record(f'{key}_central_aggregation'):
+            y_c = _agg(engine, engine.central_view, x_local, None, src_scale,
+                       dst_scale)
 
     with engine.timer.record(f'{key}_marginal_aggregation'):
         y_m = _agg(engine, engine.marginal_view, x_local, remote, src_scale,
