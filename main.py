@@ -32,6 +32,8 @@ def main():
     p.add_argument('--num_parts', type=int, default=None,
                    help='expected partition count; validated against WORLD_SIZE')
     p.add_argument('--partition_dir', type=str, default='part_data')
+    p.add_argument('--exp_dir', type=str, default='exp',
+                   help='results root (metrics/time artifacts + logs)')
     p.add_argument('--scale', type=float, default=None,
                    help='synthetic graph scale; validated against the '
                         'partition cache (omit to adopt the cached scale)')
@@ -67,7 +69,7 @@ def main():
         start = trainer.load_checkpoint(args.ckpt_path)
         print(f'resumed from {args.ckpt_path} at epoch {start}')
     best = trainer.train(start_epoch=start)
-    trainer.save()
+    trainer.save(root=args.exp_dir)
     if trainer.comm.rank == 0:
         print(f'best: epoch {best["epoch"]} val {best["val"]:.4f} '
               f'test {best["test"]:.4f}')

@@ -1,10 +1,21 @@
 #!/bin/bash
-# ogbn-products 8-part sweep over {gcn,sage} x modes (reference: scripts/*_all.sh)
+# ogbn-products sweep: {gcn,sage} x partition counts x {Vanilla,AdaQP}
+# + table (reference: scripts/products_all.sh; the 8-part GraphSAGE
+# AdaQP row is the BASELINE.json headline config). PARTS/MODES/MODELS/EXTRA env.
 set -e
-for MODEL in gcn sage; do
-  for MODE in Vanilla AdaQP; do
-    python -m torch.distributed.run --nnodes=1 --nproc-per-node 8 \
-      --master-addr 127.0.0.1 --master-port 29500 \
-      main.py --dataset ogbn-products --model_name $MODEL --mode $MODE
+cd "$(dirname "$0")/.."
+PARTS="${PARTS:-4 8}"
+MODELS="${MODELS:-gcn sage}"
+MODES="${MODES:-Vanilla AdaQP}"
+PORT="${PORT:-29503}"
+for P in $PARTS; do
+  for MODEL in $MODELS; do
+    for MODE in $MODES; do
+      python -m torch.distributed.run --nnodes=1 --nproc-per-node "$P" \
+        --master-addr 127.0.0.1 --master-port "$PORT" \
+        main.py --dataset ogbn-products --model_name "$MODEL" --mode "$MODE" \
+        --assign_scheme adaptive $EXTRA
+    done
   done
 done
+python tools/results_table.py --root "${EXP:-exp}" --dataset ogbn-products

@@ -1,0 +1,20 @@
+#!/bin/bash
+# Yelp sweep: {gcn,sage} x partition counts x {Vanilla,AdaQP} + table
+# (reference: scripts/yelp_all.sh). Quick runs: PARTS/MODES/MODELS/EXTRA env.
+set -e
+cd "$(dirname "$0")/.."
+PARTS="${PARTS:-2 4}"
+MODELS="${MODELS:-gcn sage}"
+MODES="${MODES:-Vanilla AdaQP}"
+PORT="${PORT:-29501}"
+for P in $PARTS; do
+  for MODEL in $MODELS; do
+    for MODE in $MODES; do
+      python -m torch.distributed.run --nnodes=1 --nproc-per-node "$P" \
+        --master-addr 127.0.0.1 --master-port "$PORT" \
+        main.py --dataset yelp --model_name "$MODEL" --mode "$MODE" \
+        --assign_scheme adaptive $EXTRA
+    done
+  done
+done
+python tools/results_table.py --root "${EXP:-exp}" --dataset yelp
