@@ -210,7 +210,7 @@ template <> struct RawVec<2>  { using type = ushort; };
 
 template <int SW, typename T, int VE>
 __global__ void spmm_csr_kernel(
-    const int64_t* __restrict__ indices,
+    const int32_t* __restrict__ indices,   // int32: halves index bytes
     const T* __restrict__ xl, const T* __restrict__ xr,
     T* __restrict__ y,
     const float* __restrict__ src_scale, const float* __restrict__ dst_scale,
@@ -346,24 +346,34 @@ fused_dual_gemm_bf16_kernel(
         const int64_t arow_g = row0 + arow;
         const ushort* arow_p = A + arow_g * K;
         const ushort* wt_p = Wt + arow * K;   // row (=out col) within 16-tile
-        for (int64_t k0 = 0; k0 < K; k0 += 32) {
-            const int64_t k = k0 + kgrp * 8;
-            bf16x8 af = {};
-            if (arow_g < M && k < K)               // K % 8 == 0: whole sub-block
-                af = load_bf16x8(arow_p + k);
-            bf16x8 bfr[FDG_MAXNT];
+        // double-buffered K pipeline: while the MFMAs of chunk i issue,
+        // the loads of chunk i+1 are already in flight (round-1 version
+        // had no K-pipelining — every chunk stalled on L2/HBM latency)
+        bf16x8 af0 = {}, af1 = {};
+        bf16x8 b0[FDG_MAXNT], b1[FDG_MAXNT];
+        auto load_chunk = [&](int64_t k, bf16x8& af, bf16x8* bfr) {
+            af = (arow_g < M && k < K) ? load_bf16x8(arow_p + k) : bf16x8{};
 #pragma unroll
             for (int nt = 0; nt < FDG_MAXNT; ++nt) {
                 if (nt >= NT) break;
                 bfr[nt] = (k < K) ? load_bf16x8(wt_p + (int64_t)nt * 16 * K + k)
                                   : bf16x8{};
             }
+        };
+        auto mfma_chunk = [&](const bf16x8& af, const bf16x8* bfr) {
 #pragma unroll
             for (int nt = 0; nt < FDG_MAXNT; ++nt) {
                 if (nt >= NT) break;
                 acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                     af, bfr[nt], acc[nt], 0, 0, 0);
             }
+        };
+        load_chunk(kgrp * 8, af0, b0);
+        for (int64_t k0 = 0; k0 < K; k0 += 64) {
+            if (k0 + 32 < K) load_chunk(k0 + 32 + kgrp * 8, af1, b1);
+            mfma_chunk(af0, b0);
+            if (k0 + 64 < K) load_chunk(k0 + 64 + kgrp * 8, af0, b0);
+            if (k0 + 32 < K) mfma_chunk(af1, b1);
         }
     }
 
@@ -490,7 +500,9 @@ void spmm_csr(torch::Tensor indices, torch::Tensor xl,
     int64_t blocks = (n_seg + rows_per_block - 1) / rows_per_block;
     blocks = std::max<int64_t>(std::min<int64_t>(blocks, 16384), 1);
     const dim3 grid(blocks), block(block_threads);
-    const int64_t* ind_p = indices.data_ptr<int64_t>();
+    TORCH_CHECK(indices.scalar_type() == torch::kInt32,
+                "spmm_csr expects int32 indices (SpmmView builds them)");
+    const int32_t* ind_p = indices.data_ptr<int32_t>();
     const float* ss_p = src_scale.numel() ? src_scale.data_ptr<float>() : nullptr;
     const float* ds_p = dst_scale.numel() ? dst_scale.data_ptr<float>() : nullptr;
     const int32_t* sr_p = seg_row.data_ptr<int32_t>();
