@@ -96,6 +96,10 @@ class Communicator:
         GPU, or forcing the reference's CPU-staged transport) stages
         through host memory like the reference's pinned-CPU path
         (``comm.py:173-189``)."""
+        if out.numel() == 0 and inp.numel() == 0:
+            # fully isolated partitions: nothing to move. Skipping keeps
+            # the nccl path away from 0-numel device collectives.
+            return None
         if out.is_cuda and 'nccl' not in self.backend:
             inp_c = inp.cpu()
             out_c = torch.empty(out.shape, dtype=out.dtype)
