@@ -53,27 +53,31 @@ def edge_cut(assign: Tensor, src: Tensor, dst: Tensor) -> int:
 
 def grow_assignment(g: GlobalGraph, num_parts: int, seed: int = 0,
                     imbalance: float = 0.05, refine_passes: int = 100,
+                    coarsen_above: int = 100_000,
                     device: Optional[torch.device] = None) -> Tensor:
-    """Balanced multi-seed region-growing partitioner for ARBITRARY
-    graphs — the METIS stand-in (reference uses
-    ``dgl.distributed.partition_graph``, ``helper/partition.py:70-72``).
+    """Balanced partitioner for ARBITRARY graphs — the METIS stand-in
+    (reference uses ``dgl.distributed.partition_graph``,
+    ``helper/partition.py:70-72``). Fully vectorized; runs on GPU when
+    available (full ogbn-products, 2.4M nodes / 126M directed edges, in
+    seconds).
 
-    Fully vectorized (frontier tensors, no Python node loop — the round-1
-    version walked nodes one by one and could not touch 2.4M nodes):
+    Small graphs (n <= ``coarsen_above``): multi-seed region growing —
+    all parts expand level-synchronously; each round every unassigned
+    node adjacent to an assigned one joins the open part owning most of
+    its neighbors (one bincount over n*P keys), capacity enforced by
+    keeping the highest-affinity claimants. Then greedy KL-lite boundary
+    refinement to convergence.
 
-    1. P random seeds; all parts grow level-synchronously. Each round,
-       every unassigned node adjacent to an assigned one counts its
-       neighbors per part (one bincount over n*P keys) and joins the
-       part with the most neighbors among parts with remaining quota;
-       per-part capacity is enforced by keeping the highest-count
-       claimants (sort + rank-within-part).
-    2. Disconnected leftovers are seeded into the emptiest parts.
-    3. ``refine_passes`` greedy boundary-refinement sweeps: boundary
-       nodes move to their neighbor-majority part when the gain is
-       positive and balance (±``imbalance``) allows, best gains first.
-
-    O(E) tensor work per round; rounds ≈ graph diameter. Runs on GPU
-    when available (2.4M nodes / 250M edges in seconds)."""
+    Large graphs: MULTILEVEL — plain growth degenerates at millions of
+    nodes (the frontier floods the graph in ~2 rounds of near-random
+    tie-breaks and local refinement cannot repair macro-structure;
+    measured cut 0.49 vs 0.098 planted at full products scale). So:
+    (1) label-propagation clustering collapses dense regions (capped at
+    quota/4 so packing stays feasible), (2) the weighted contracted
+    graph (cluster sizes as node weights, multiplicities as edge
+    weights) is partitioned by weighted region growing seeded at the
+    heaviest clusters, (3) the projection is polished by the same
+    boundary refinement on the full graph."""
     n, P = g.num_nodes, num_parts
     if P <= 1:
         return torch.zeros(n, dtype=torch.int64)
@@ -85,15 +89,74 @@ def grow_assignment(g: GlobalGraph, num_parts: int, seed: int = 0,
     noself = src != dst
     src, dst = src[noself], dst[noself]
 
-    quota = torch.full((P,), (n + P - 1) // P, dtype=torch.int64,
+    if n > coarsen_above:
+        labels, C = _label_prop_clusters(src, dst, n, cap=max(n // (P * 4), 1))
+        cs, cd = labels[src], labels[dst]
+        m = cs != cd
+        key = cd[m] * C + cs[m]
+        uk, w = torch.unique(key, return_counts=True)
+        csrc, cdst, ew = uk % C, uk // C, w.float()
+        node_w = torch.bincount(labels, minlength=C)
+        seeds = torch.topk(node_w, min(P, C)).indices
+        cassign = _grow_core(csrc, cdst, C, P, node_w, ew, seeds, gen, device)
+        assign = cassign[labels]
+    else:
+        node_w = torch.ones(n, dtype=torch.int64, device=device)
+        seeds = torch.randperm(n, generator=gen)[:P].to(device)
+        assign = _grow_core(src, dst, n, P, node_w, None, seeds, gen, device)
+
+    assign = _refine(assign, src, dst, n, P, imbalance, refine_passes)
+    return assign.cpu()
+
+
+def _label_prop_clusters(src: Tensor, dst: Tensor, n: int, cap: int,
+                         rounds: int = 6) -> Tuple[Tensor, int]:
+    """Weighted-majority label propagation (synchronous, self-vote to
+    damp oscillation), then oversized clusters are split into <=cap
+    chunks so the packing stage stays feasible. Returns (labels in
+    [0,C), C)."""
+    device = src.device
+    labels = torch.arange(n, device=device)
+    self_ix = torch.arange(n, device=device)
+    for _ in range(rounds):
+        key = torch.cat([dst * n + labels[src], self_ix * n + labels])
+        key, _ = torch.sort(key)
+        uk, counts = torch.unique_consecutive(key, return_counts=True)
+        d = uk // n
+        lab = uk % n
+        # prefer higher count, then smaller label (deterministic)
+        comb = counts * n + (n - 1 - lab)
+        best = torch.zeros(n, dtype=torch.int64, device=device)
+        best.scatter_reduce_(0, d, comb, reduce='amax', include_self=True)
+        labels = torch.where(best > 0, (n - 1) - best % n, labels)
+    _, labels = torch.unique(labels, return_inverse=True)
+    C = int(labels.max()) + 1
+    # split clusters larger than cap (rank-chunking members)
+    sizes = torch.bincount(labels, minlength=C)
+    if int(sizes.max()) > cap:
+        rank = _rank_within(labels, C)
+        labels = labels + C * (rank // cap)
+        _, labels = torch.unique(labels, return_inverse=True)
+        C = int(labels.max()) + 1
+    return labels, C
+
+
+def _grow_core(src: Tensor, dst: Tensor, n: int, P: int, node_w: Tensor,
+               edge_w: Optional[Tensor], seeds: Tensor,
+               gen: torch.Generator, device) -> Tensor:
+    """Weighted multi-seed region growing. Balance is on SUM of node_w
+    per part; affinity counts are edge_w-weighted. Termination: any
+    zero-progress round relaxes the quota."""
+    total = int(node_w.sum())
+    quota = torch.full((P,), (total + P - 1) // P, dtype=torch.int64,
                        device=device)
     assign = torch.full((n,), -1, dtype=torch.int64, device=device)
-    seeds = torch.randperm(n, generator=gen)[:P].to(device)
-    assign[seeds] = torch.arange(P, device=device)
-    sizes = torch.bincount(assign[assign >= 0], minlength=P)
+    assign[seeds] = torch.arange(seeds.numel(), device=device)
+    sizes = torch.zeros(P, dtype=torch.int64, device=device)
+    sizes.scatter_add_(0, assign[seeds], node_w[seeds])
+    bump = max(total // (P * 50), 1)
 
     while True:
-        # candidate nodes: unassigned with >=1 assigned in-neighbor
         e = (assign[src] >= 0) & (assign[dst] < 0)
         if not bool(e.any()):
             left = torch.nonzero(assign < 0, as_tuple=True)[0]
@@ -104,43 +167,53 @@ def grow_assignment(g: GlobalGraph, num_parts: int, seed: int = 0,
             order = torch.argsort(sizes)[:k]
             pick = left[torch.randperm(left.numel(), device=device)[:k]]
             assign[pick] = order
-            sizes = sizes.scatter_add(0, order, torch.ones_like(order))
+            sizes.scatter_add_(0, order, node_w[pick])
             continue
         es, ed = src[e], dst[e]
         cand = torch.unique(ed)
-        counts = torch.bincount(ed * P + assign[es],
-                                minlength=n * P).view(n, P)[cand]
+        wts = edge_w[e] if edge_w is not None else None
+        counts = torch.bincount(ed * P + assign[es], weights=wts,
+                                minlength=n * P).view(n, P)[cand].float()
         open_parts = sizes < quota
         if not bool(open_parts.any()):
-            quota = quota + 1          # all full but nodes remain
+            quota = quota + bump
             open_parts = sizes < quota
-        masked = counts.float()
-        masked[:, ~open_parts] = -1.0
-        best_cnt, best_part = masked.max(dim=1)
+        counts[:, ~open_parts] = -1.0
+        best_cnt, best_part = counts.max(dim=1)
         ok = best_cnt > 0
         cand, best_cnt, best_part = cand[ok], best_cnt[ok], best_part[ok]
         if cand.numel() == 0:
-            # every candidate's neighbors sit in full parts: relax quota
-            # so the next round makes progress (guaranteed termination)
-            quota = quota + 1
+            quota = quota + bump   # progress guarantee
             continue
-        # capacity: keep the highest-count claimants per part
-        order = torch.argsort(best_part.float() * float(n + 1) - best_cnt)
-        cand, best_cnt, best_part = cand[order], best_cnt[order], best_part[order]
-        claims = torch.bincount(best_part, minlength=P)
-        start = torch.cumsum(claims, 0) - claims
-        rank = torch.arange(cand.numel(), device=device) - start[best_part]
-        keep = rank < (quota - sizes)[best_part]
+        # capacity: within each part, admit highest-affinity claimants
+        # while their cumulative weight fits the remaining quota
+        order = torch.argsort(best_part.float() * (best_cnt.max() + 1.0)
+                              - best_cnt)
+        cand, best_part = cand[order], best_part[order]
+        w = node_w[cand]
+        cw = torch.cumsum(w, 0)
+        claims_w = torch.zeros(P, dtype=torch.int64, device=device)
+        claims_w.scatter_add_(0, best_part, w)
+        starts = torch.cumsum(claims_w, 0) - claims_w   # weight before part
+        within = cw - w - starts[best_part]             # weight admitted before me
+        keep = within < (quota - sizes)[best_part]
         assign[cand[keep]] = best_part[keep]
-        sizes = sizes + torch.bincount(best_part[keep], minlength=P)
+        sizes.scatter_add_(0, best_part[keep], w[keep])
+    return assign
 
-    # greedy boundary refinement (KL-lite, vectorized); runs until the
-    # cut stops improving — converges to planted-cluster quality on
-    # scrambled synthetic graphs (tests/test_partitioner.py)
+
+def _refine(assign: Tensor, src: Tensor, dst: Tensor, n: int, P: int,
+            imbalance: float, passes: int) -> Tensor:
+    """Greedy KL-lite boundary refinement (vectorized): boundary nodes
+    move to their neighbor-majority part when gain > 0 and balance
+    (±imbalance) allows, best gains first; stops when the cut stops
+    improving."""
+    device = src.device
+    sizes = torch.bincount(assign, minlength=P)
     hi = int((n / P) * (1 + imbalance)) + 1
     lo = int((n / P) * (1 - imbalance))
     best_cut, stall = None, 0
-    for _ in range(refine_passes):
+    for _ in range(passes):
         counts = torch.bincount(dst * P + assign[src], minlength=n * P)
         counts = counts.view(n, P)
         cur = counts.gather(1, assign[:, None]).squeeze(1)
@@ -158,7 +231,6 @@ def grow_assignment(g: GlobalGraph, num_parts: int, seed: int = 0,
             break
         nodes = torch.nonzero(mov, as_tuple=True)[0]
         # best gains first; cap by destination headroom and source floor
-        # so balance is preserved
         order = torch.argsort(gain[nodes], descending=True)
         nodes = nodes[order]
         tgt = best_part[nodes]
@@ -173,7 +245,7 @@ def grow_assignment(g: GlobalGraph, num_parts: int, seed: int = 0,
             break
         assign[nodes] = tgt
         sizes = torch.bincount(assign, minlength=P)
-    return assign.cpu()
+    return assign
 
 
 def _rank_within(groups: Tensor, P: int) -> Tensor:

@@ -30,12 +30,19 @@ def bench_mixed():
             rows_pp.append(torch.randperm(per * 2, generator=g)[:per] + base)
             base += per * 2
         plan = _layout(bits_pp, rows_pp, F).to('cuda')
+        # recv side: same bit mix, but rows index the REMOTE block
+        # (arange blocks, like build_key_plan's recv side) — using the
+        # send rows here would scatter out of bounds
+        rbase, rrows = 0, []
+        for p in range(peers):
+            rrows.append(torch.arange(rbase, rbase + per, dtype=torch.int64))
+            rbase += per
+        rplan = _layout(bits_pp, rrows, F).to('cuda')
         x = torch.randn(base, F, device='cuda')
-        out = torch.zeros(plan.total_nodes, F, device='cuda')
-        # recv side reuses the same layout (symmetric channels)
+        out = torch.zeros(rplan.total_nodes, F, device='cuda')
         for _ in range(3):
             payload, params = mixed_quantize(x, plan, 3)
-            mixed_dequantize(payload, params, plan, out)
+            mixed_dequantize(payload, params, rplan, out)
         torch.cuda.synchronize(); t0 = time.perf_counter()
         iters = 20
         for i in range(iters):
@@ -43,7 +50,7 @@ def bench_mixed():
         torch.cuda.synchronize(); tq = (time.perf_counter() - t0) / iters
         torch.cuda.synchronize(); t0 = time.perf_counter()
         for i in range(iters):
-            mixed_dequantize(payload, params, plan, out)
+            mixed_dequantize(payload, params, rplan, out)
         torch.cuda.synchronize(); td = (time.perf_counter() - t0) / iters
         fp32_bytes = plan.total_nodes * F * 4
         wire = plan.total_bytes + 4 * plan.total_nodes   # payload + bf16 params
