@@ -101,14 +101,38 @@ class Communicator:
             # the nccl path away from 0-numel device collectives.
             return None
         if out.is_cuda and 'nccl' not in self.backend:
-            inp_c = inp.cpu()
-            out_c = torch.empty(out.shape, dtype=out.dtype)
+            # PINNED staging with stream-scoped non-blocking copies.
+            # A pageable `.cpu()` here would take the legacy null-stream
+            # path and wait for ALL streams — including the central
+            # aggregation the decomposed path wants to overlap with —
+            # serializing the very work this transport runs under
+            # (measured: AdaQP-p 145 ms vs Vanilla 128 ms before;
+            # reference keeps pinned buffers for the same reason,
+            # ``comm.py:173-189``).
+            inp_c = self._pinned('a2a_send', inp.shape, inp.dtype)
+            out_c = self._pinned('a2a_recv', out.shape, out.dtype)
+            inp_c.copy_(inp, non_blocking=True)
+            torch.cuda.current_stream().synchronize()
             dist.all_to_all_single(out_c, inp_c, list(out_splits),
                                    list(in_splits))
-            out.copy_(out_c, non_blocking=False)
+            out.copy_(out_c, non_blocking=True)
             return None
         return dist.all_to_all_single(out, inp, list(out_splits),
                                       list(in_splits), async_op=async_op)
+
+    def _pinned(self, tag: str, shape, dtype) -> Tensor:
+        """Cached page-locked staging buffer, grown as needed."""
+        cache = getattr(self, '_pin_cache', None)
+        if cache is None:
+            cache = self._pin_cache = {}
+        need = 1
+        for s in shape:
+            need *= int(s)
+        buf = cache.get((tag, dtype))
+        if buf is None or buf.numel() < need:
+            buf = torch.empty(max(need, 1), dtype=dtype, pin_memory=True)
+            cache[(tag, dtype)] = buf
+        return buf[:need].view(shape)
 
     def exchange_rows(self, send: Tensor, send_splits: Sequence[int],
                       recv_splits: Sequence[int], out: Optional[Tensor] = None,
