@@ -69,7 +69,11 @@ def worker(rank, world, port, mode, scale, dataset, epochs, warmup, dtype, q):
         el = torch.tensor([time.perf_counter() - t0])
         comm.all_reduce_max(el)
         if rank == 0:
-            q.put((mode, float(el.item()) / epochs * 1e3))
+            ms = float(el.item()) / epochs * 1e3
+            if q is None:
+                print(f'{mode}: {ms:.1f} ms/epoch', flush=True)
+            else:
+                q.put((mode, ms))
     finally:
         Communicator.shutdown()
 
@@ -84,7 +88,17 @@ def main():
     p.add_argument('--warmup', type=int, default=3)
     p.add_argument('--dtype', default='fp32', choices=['fp32', 'bf16'])
     p.add_argument('--out', default=None)
+    p.add_argument('--worker', type=int, default=None,
+                   help='run ONE rank in-process (no mp.spawn) so a '
+                        'profiler can wrap a single worker; launch the '
+                        'peer rank separately with the other index')
+    p.add_argument('--port', type=int, default=29640)
     args = p.parse_args()
+    if args.worker is not None:
+        mode = args.mode if args.mode != 'all' else 'AdaQP-p'
+        worker(args.worker, 2, args.port, mode, args.scale, args.dataset,
+               args.epochs, args.warmup, args.dtype, None)
+        return
     modes = MODES if args.mode == 'all' else [args.mode]
     ctx = mp.get_context('spawn')
     results = {}
