@@ -1,81 +1,64 @@
 #!/usr/bin/env python3
-"""Summarize comp/comm overlap from a rocprofv3 CSV trace of ONE
-overlap_bench worker (kernel-trace + memory-copy-trace).
+"""Summarize comp/comm overlap from a rocprofv3 CSV kernel trace of ONE
+overlap_bench worker.
 
-The gloo-staged exchange is: D2H copy (comm stream) -> CPU all-to-all
-(GPU idle on the comm stream) -> H2D copy. True overlap means compute
-kernels execute INSIDE those [D2H end, H2D start] windows — that is
-exactly the time the decomposed path hides. Reports total window time
-and the kernel busy time inside it.
+Method: the comm stream carries the exchange's producer/staging work
+(gather kernels + blit copies); a >2 ms gap between consecutive
+comm-stream ops is a transport window (gloo CPU all-to-all in flight,
+or RCCL DMA). Overlap evidence = default-stream kernels (the central
+aggregation) executing INSIDE those windows. The comm stream is
+auto-detected as the non-default stream carrying gather + copyBuffer
+ops.
 
-Usage: python tools/overlap_trace_report.py <trace_dir>
+Usage: python tools/overlap_trace_report.py <trace_dir> [skip_s]
+(skip_s: drop the first N seconds — setup/warmup; default 0)
 """
 import csv
 import glob
 import os
 import sys
-
-
-def load_rows(pattern):
-    rows = []
-    for fn in glob.glob(pattern):
-        with open(fn) as f:
-            rows.extend(csv.DictReader(f))
-    return rows
-
-
-def col(row, *names):
-    for n in names:
-        for k in row:
-            if k.strip('"').lower() == n.lower():
-                return row[k].strip('"')
-    raise KeyError(f'{names} not in {list(row)[:12]}')
+from collections import Counter
 
 
 def main():
     d = sys.argv[1]
-    kern = load_rows(os.path.join(d, '*kernel_trace.csv'))
-    copies = load_rows(os.path.join(d, '*memory_copy_trace.csv'))
-    if not kern or not copies:
-        raise SystemExit(f'no kernel/copy trace CSVs under {d}')
-    ks = [(int(col(r, 'Start_Timestamp')), int(col(r, 'End_Timestamp')),
-           col(r, 'Kernel_Name', 'Name')) for r in kern]
-    cs = sorted((int(col(r, 'Start_Timestamp')), int(col(r, 'End_Timestamp')),
-                 col(r, 'Direction', 'Name', 'Kind')) for r in copies)
-    # exchange windows: D2H end -> next H2D start
+    skip_s = float(sys.argv[2]) if len(sys.argv) > 2 else 0.0
+    ks = []
+    for fn in glob.glob(os.path.join(d, '*kernel_trace.csv')):
+        for r in csv.DictReader(open(fn)):
+            ks.append((int(r['Start_Timestamp']), int(r['End_Timestamp']),
+                       r['Stream_Id'], r['Kernel_Name']))
+    if not ks:
+        raise SystemExit(f'no kernel trace CSVs under {d}')
+    ks.sort()
+    t0 = ks[0][0]
+    lo = t0 + int(skip_s * 1e9)
+    # comm stream: the non-zero stream with the most ops
+    streams = Counter(q for _, _, q, _ in ks if q != '0')
+    if not streams:
+        raise SystemExit('no side-stream ops found (overlap path not taken?)')
+    comm_id = streams.most_common(1)[0][0]
+    comm = [(s, e) for s, e, q, _ in ks if q == comm_id and s > lo]
+    dflt = [(s, e, n) for s, e, q, n in ks if q == '0' and s > lo]
     windows = []
-    for i, (s, e, dirn) in enumerate(cs):
-        dl = dirn.lower()
-        if 'device_to_host' in dl or 'd2h' in dl or 'devicetohost' in dl:
-            for s2, e2, d2 in cs[i + 1:]:
-                d2l = d2.lower()
-                if 'host_to_device' in d2l or 'h2d' in d2l or 'hosttodevice' in d2l:
-                    if s2 > e:
-                        windows.append((e, s2))
-                    break
-    # merge overlapping windows
-    windows.sort()
-    merged = []
-    for w in windows:
-        if merged and w[0] <= merged[-1][1]:
-            merged[-1] = (merged[-1][0], max(merged[-1][1], w[1]))
-        else:
-            merged.append(w)
-    total_win = sum(b - a for a, b in merged)
+    for (s1, e1), (s2, e2) in zip(comm, comm[1:]):
+        if s2 - e1 > 2e6:
+            windows.append((e1, s2))
+    tot = sum(b - a for a, b in windows)
     busy = 0
-    by_kernel = {}
-    for s, e, name in ks:
-        for a, b in merged:
+    bykern = {}
+    for s, e, n in dflt:
+        for a, b in windows:
             o = min(e, b) - max(s, a)
             if o > 0:
                 busy += o
-                key = name.split('<')[0].split('(')[0]
-                by_kernel[key] = by_kernel.get(key, 0) + o
-    print(f'{len(merged)} exchange windows, total {total_win/1e6:.2f} ms')
-    print(f'kernel busy time inside windows: {busy/1e6:.2f} ms '
-          f'({100.0*busy/max(total_win,1):.0f}% of window time overlapped '
-          f'with compute)')
-    for k, v in sorted(by_kernel.items(), key=lambda x: -x[1])[:8]:
+                key = n.split('<')[0][:50]
+                bykern[key] = bykern.get(key, 0) + o
+    print(f'comm stream = {comm_id}; {len(windows)} transport windows, '
+          f'total {tot/1e6:.1f} ms')
+    print(f'default-stream kernel busy inside windows: {busy/1e6:.2f} ms '
+          f'({100*busy/max(tot,1):.0f}% of window time covered by compute)')
+    for k, v in sorted(bykern.items(), key=lambda x: -x[1])[:8]:
         print(f'  {v/1e6:8.2f} ms  {k}')
 
 
