@@ -116,12 +116,13 @@ def _scales(engine, mode: PropagationMode) -> Tuple[Optional[Tensor], Optional[T
     raise ValueError(f'unknown aggregator {engine.agg_type}')
 
 
-def _agg(engine, view, x_local: Tensor, x_remote, src_scale, dst_scale) -> Tensor:
+def _agg(engine, view, x_local: Tensor, x_remote, src_scale, dst_scale,
+         out=None) -> Tensor:
     """SpMM over one SpmmView. src_scale is the full [N] vector; dst_scale
     is sliced to the view's row range."""
     ds = (dst_scale[view.base:view.base + view.nrows]
           if dst_scale is not None else None)
-    return spmm(view, x_local, x_remote, src_scale, ds)
+    return spmm(view, x_local, x_remote, src_scale, ds, out=out)
 
 
 def _self_term(engine, x_local: Tensor, mode: PropagationMode) -> Tensor:
@@ -172,26 +173,34 @@ def decomposed_propagation(engine, x_local: Tensor, key: str, is_train: bool,
         x_local.record_stream(engine.comm_stream)
         with torch.cuda.stream(engine.comm_stream):
             staged = _exchange_start(engine, x_local, key, quant)
+        # central and marginal SpMMs write disjoint row slices of ONE
+        # output tensor (no torch.cat: the cat was a full [I,F]
+        # read+write per propagation)
+        C = g.num_central
+        y = torch.empty(g.num_inner, x_local.shape[1], dtype=x_local.dtype,
+                        device=x_local.device)
         # central rows only touch local columns -> enqueue NOW, before
         # the transport can block the host
         with engine.timer.record(f'{key}_central_aggregation'):
-            y_c = _agg(engine, engine.central_view, x_local, None, src_scale,
-                       dst_scale)
+            _agg(engine, engine.central_view, x_local, None, src_scale,
+                 dst_scale, out=y[:C])
         with torch.cuda.stream(engine.comm_stream):
             remote = _exchange_finish(engine, staged, key)
             engine.remote_ready.record(engine.comm_stream)
         torch.cuda.current_stream().wait_event(engine.remote_ready)
         remote.record_stream(torch.cuda.current_stream())
+        with engine.timer.record(f'{key}_marginal_aggregation'):
+            _agg(engine, engine.marginal_view, x_local, remote, src_scale,
+                 dst_scale, out=y[C:])
     else:
         remote = _exchange(engine, x_local, key, is_train)
         with engine.timer.record(f'{key}_central_aggregation'):
             y_c = _agg(engine, engine.central_view, x_local, None, src_scale,
                        dst_scale)
-
-    with engine.timer.record(f'{key}_marginal_aggregation'):
-        y_m = _agg(engine, engine.marginal_view, x_local, remote, src_scale,
-                   dst_scale)
-    y = torch.cat([y_c, y_m], dim=0)
+        with engine.timer.record(f'{key}_marginal_aggregation'):
+            y_m = _agg(engine, engine.marginal_view, x_local, remote,
+                       src_scale, dst_scale)
+        y = torch.cat([y_c, y_m], dim=0)
     if add_self:
         y = y + _self_term(engine, x_local, mode)
     return y

@@ -192,12 +192,16 @@ class SpmmView:
 
 
 def spmm(view: 'SpmmView', x_local: Tensor, x_remote: Optional[Tensor],
-         src_scale: Optional[Tensor], dst_scale: Optional[Tensor]) -> Tensor:
+         src_scale: Optional[Tensor], dst_scale: Optional[Tensor],
+         out: Optional[Tensor] = None) -> Tensor:
     """Aggregation SpMM over an in-edge CSR view.
 
     Columns < len(x_local) read x_local; the rest read x_remote (the
     all-to-all output block) — no concat on the GPU path.
     src_scale: [n_local+n_remote] or None; dst_scale: [nrows] or None.
+    ``out`` (GPU path): write into this [nrows, F] contiguous tensor —
+    lets the decomposed path target row slices of ONE output so the
+    central/marginal results need no torch.cat afterwards.
     """
     indptr, indices, num_rows = view.indptr, view.indices, view.nrows
     if not x_local.is_cuda and view.col_blocked:
@@ -205,8 +209,11 @@ def spmm(view: 'SpmmView', x_local: Tensor, x_remote: Optional[Tensor],
                            'fallback consumes indptr, which the blocked '
                            'layout does not preserve)')
     if x_local.is_cuda:
-        out = torch.empty(num_rows, x_local.shape[1], dtype=x_local.dtype,
-                          device=x_local.device)
+        if out is None:
+            out = torch.empty(num_rows, x_local.shape[1], dtype=x_local.dtype,
+                              device=x_local.device)
+        else:
+            assert out.is_contiguous() and out.shape[0] == num_rows
         empty = torch.empty(0, device=x_local.device)
         native().spmm_csr(indices, x_local,
                           x_remote if x_remote is not None else empty, out,
