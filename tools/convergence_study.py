@@ -15,7 +15,8 @@ import torch
 import torch.multiprocessing as mp
 
 
-def worker(rank, world, port, mode, scheme, epochs, q):
+def worker(rank, world, port, mode, scheme, epochs, q, nodes=4000,
+           edges=48000, use_gpu=False):
     os.environ.update(MASTER_ADDR='127.0.0.1', MASTER_PORT=str(port),
                       RANK=str(rank), WORLD_SIZE=str(world))
     from adaqp_amd.comm import Communicator
@@ -28,17 +29,21 @@ def worker(rank, world, port, mode, scheme, epochs, q):
     comm = Communicator(backend='gloo')
     try:
         torch.manual_seed(11)
-        g = random_partitioned_graph(4000, 48000, 32, 8, world, seed=21,
+        dev = torch.device('cuda:0') if use_gpu else torch.device('cpu')
+        if use_gpu:
+            torch.cuda.set_device(dev)
+            comm.device = dev
+        g = random_partitioned_graph(nodes, edges, 32, 8, world, seed=21,
                                      cut_frac=0.3, teacher_labels=True)
         lg = partition_all(g, world)[rank]
         engine = GraphEngine(lg, RunMode(mode), DistGNNType.DistGCN,
-                             msg_dims=[32, 64, 64], device=torch.device('cpu'))
+                             msg_dims=[32, 64, 64], device=dev)
         assigner = Assigner(engine, AssignScheme(scheme), group_size=100,
                             init_bits=8)
         if engine.bit_type.name == 'QUANT':
             assigner.initial_assignment()
         torch.manual_seed(33)
-        model = DistGCN(32, 64, 8, num_layers=3, dropout=0.0)
+        model = DistGCN(32, 64, 8, num_layers=3, dropout=0.0).to(dev)
         comm.sync_model_params(model)
         opt = torch.optim.Adam(model.parameters(), lr=0.01)
         gc = global_train_count(engine)
@@ -57,11 +62,12 @@ def worker(rank, world, port, mode, scheme, epochs, q):
         Communicator.shutdown()
 
 
-def run(mode, scheme, epochs, port):
+def run(mode, scheme, epochs, port, nodes=4000, edges=48000, use_gpu=False):
     ctx = mp.get_context('spawn')
     q = ctx.SimpleQueue()
     procs = [ctx.Process(target=worker, args=(r, 2, port, mode, scheme,
-                                              epochs, q))
+                                              epochs, q, nodes, edges,
+                                              use_gpu))
              for r in range(2)]
     for p in procs:
         p.start()
@@ -74,12 +80,19 @@ def run(mode, scheme, epochs, port):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument('--epochs', type=int, default=200)
+    ap.add_argument('--nodes', type=int, default=4000)
+    ap.add_argument('--edges', type=int, default=48000)
+    ap.add_argument('--gpu', action='store_true',
+                    help='2 ranks sharing cuda:0 (gloo-staged transport) — '
+                         'runs the FULL adaptive pipeline on the HIP '
+                         'kernels at larger scale')
     args = ap.parse_args()
     results = {}
     for i, (mode, scheme) in enumerate([('Vanilla', 'uniform'),
                                         ('AdaQP', 'adaptive'),
                                         ('AdaQP-q', 'uniform')]):
-        m, s, curve = run(mode, scheme, args.epochs, 29720 + i)
+        m, s, curve = run(mode, scheme, args.epochs, 29720 + i,
+                          args.nodes, args.edges, args.gpu)
         results[f'{m}/{s}'] = curve
         print(f'{m}/{s}: final test acc {curve[-1]:.4f} '
               f'(best {max(curve):.4f})')
