@@ -238,4 +238,7 @@ def spmm(view: 'SpmmView', x_local: Tensor, x_remote: Optional[Tensor],
     y = torch.sparse.mm(sp, xs.float()).to(x.dtype)
     if dst_scale is not None:
         y = y * dst_scale[:, None]
+    if out is not None:
+        out.copy_(y)
+        return out
     return y
