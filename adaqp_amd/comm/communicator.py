@@ -130,7 +130,10 @@ class Communicator:
             need *= int(s)
         buf = cache.get((tag, dtype))
         if buf is None or buf.numel() < need:
-            buf = torch.empty(max(need, 1), dtype=dtype, pin_memory=True)
+            # pinning requires a CUDA context; the staging path only runs
+            # with CUDA tensors, but keep the helper usable on CPU hosts
+            buf = torch.empty(max(need, 1), dtype=dtype,
+                              pin_memory=torch.cuda.is_available())
             cache[(tag, dtype)] = buf
         return buf[:need].view(shape)
 
