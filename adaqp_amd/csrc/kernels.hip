@@ -214,8 +214,8 @@ __global__ void spmm_csr_kernel(
     const T* __restrict__ xl, const T* __restrict__ xr,
     T* __restrict__ y,
     const float* __restrict__ src_scale, const float* __restrict__ dst_scale,
-    const int32_t* __restrict__ seg_row, const int64_t* __restrict__ seg_e0,
-    const int64_t* __restrict__ seg_e1, const uint8_t* __restrict__ seg_multi,
+    const int32_t* __restrict__ seg_row, const int32_t* __restrict__ seg_e0,
+    const int32_t* __restrict__ seg_e1, const uint8_t* __restrict__ seg_multi,
     int64_t n_seg, int64_t F, int64_t n_local) {
     constexpr bool BF = sizeof(T) == 2;
     constexpr int NB = VE * sizeof(T);
@@ -506,8 +506,11 @@ void spmm_csr(torch::Tensor indices, torch::Tensor xl,
     const float* ss_p = src_scale.numel() ? src_scale.data_ptr<float>() : nullptr;
     const float* ds_p = dst_scale.numel() ? dst_scale.data_ptr<float>() : nullptr;
     const int32_t* sr_p = seg_row.data_ptr<int32_t>();
-    const int64_t* e0_p = seg_e0.data_ptr<int64_t>();
-    const int64_t* e1_p = seg_e1.data_ptr<int64_t>();
+    TORCH_CHECK(seg_e0.scalar_type() == torch::kInt32 &&
+                seg_e1.scalar_type() == torch::kInt32,
+                "seg_e0/e1 must be int32 (SpmmView builds them)");
+    const int32_t* e0_p = seg_e0.data_ptr<int32_t>();
+    const int32_t* e1_p = seg_e1.data_ptr<int32_t>();
     const uint8_t* sm_p = seg_multi.data_ptr<uint8_t>();
     auto run = [&](auto sw_tag, auto t_tag, auto ve_tag) {
         constexpr int SWC = decltype(sw_tag)::value;

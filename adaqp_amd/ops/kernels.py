@@ -179,10 +179,12 @@ class SpmmView:
             self.seg_e1 = e1
             self.seg_multi = multi_mask[seg_row].to(torch.uint8)
             self.zero_rows = torch.nonzero(multi_mask, as_tuple=True)[0].to(torch.int32)
-        # int32 column indices: halves the kernel's index-read bytes
-        # (~1 GB -> 0.5 GB per pass on full products); node counts are
-        # always < 2^31
+        # int32 column indices + segment bounds: halves the kernel's
+        # index-read bytes (~1 GB -> 0.5 GB per pass on full products);
+        # node/edge counts are always < 2^31
         self.indices = self.indices.to(torch.int32)
+        self.seg_e0 = self.seg_e0.to(torch.int32)
+        self.seg_e1 = self.seg_e1.to(torch.int32)
         self._on = None
 
     def to(self, device):
