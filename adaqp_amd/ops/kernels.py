@@ -106,7 +106,9 @@ def mixed_dequantize(payload: Tensor, params: Tensor, plan: SidePlan,
 
 _csr_cache: dict = {}
 
-SEG_EDGES = 512      # max edges one sub-wavefront processes serially
+import os as _os
+SEG_EDGES = int(_os.environ.get('ADAQP_SEG_EDGES', '512'))
+# max edges one sub-wavefront processes serially (env-tunable for A/B)
 
 
 class SpmmView:
