@@ -240,6 +240,36 @@ __global__ void spmm_csr_kernel(
 #pragma unroll
             for (int k = 0; k < VE; ++k) { acc0[k] = 0.f; acc1[k] = 0.f; }
             int64_t e = e0;
+            // FOUR edges in flight: the gather is latency-bound on L2
+            // misses; independent loads + accumulator pairs keep more
+            // of them outstanding (round-1 version had two)
+            for (; e + 3 < e1; e += 4) {
+                const int64_t c0 = indices[e],     c1 = indices[e + 1];
+                const int64_t c2 = indices[e + 2], c3 = indices[e + 3];
+                const float s0 = src_scale ? src_scale[c0] : 1.f;
+                const float s1 = src_scale ? src_scale[c1] : 1.f;
+                const float s2 = src_scale ? src_scale[c2] : 1.f;
+                const float s3 = src_scale ? src_scale[c3] : 1.f;
+                const T* p0 = (c0 < n_local ? xl + c0 * F : xr + (c0 - n_local) * F) + f0;
+                const T* p1 = (c1 < n_local ? xl + c1 * F : xr + (c1 - n_local) * F) + f0;
+                const T* p2 = (c2 < n_local ? xl + c2 * F : xr + (c2 - n_local) * F) + f0;
+                const T* p3 = (c3 < n_local ? xl + c3 * F : xr + (c3 - n_local) * F) + f0;
+                const Raw r0 = *reinterpret_cast<const Raw*>(p0);
+                const Raw r1 = *reinterpret_cast<const Raw*>(p1);
+                const Raw r2 = *reinterpret_cast<const Raw*>(p2);
+                const Raw r3 = *reinterpret_cast<const Raw*>(p3);
+                const T* v0 = reinterpret_cast<const T*>(&r0);
+                const T* v1 = reinterpret_cast<const T*>(&r1);
+                const T* v2 = reinterpret_cast<const T*>(&r2);
+                const T* v3 = reinterpret_cast<const T*>(&r3);
+#pragma unroll
+                for (int k = 0; k < VE; ++k) {
+                    acc0[k] = fmaf(to_f32<T>(v0[k]), s0, acc0[k]);
+                    acc1[k] = fmaf(to_f32<T>(v1[k]), s1, acc1[k]);
+                    acc0[k] = fmaf(to_f32<T>(v2[k]), s2, acc0[k]);
+                    acc1[k] = fmaf(to_f32<T>(v3[k]), s3, acc1[k]);
+                }
+            }
             for (; e + 1 < e1; e += 2) {   // VE divides F: loads always in-bounds
                 const int64_t c0 = indices[e];
                 const int64_t c1 = indices[e + 1];
