@@ -107,7 +107,7 @@ def mixed_dequantize(payload: Tensor, params: Tensor, plan: SidePlan,
 _csr_cache: dict = {}
 
 import os as _os
-SEG_EDGES = int(_os.environ.get('ADAQP_SEG_EDGES', '512'))
+SEG_EDGES = int(_os.environ.get('ADAQP_SEG_EDGES', '256'))
 # max edges one sub-wavefront processes serially (env-tunable for A/B)
 
 
