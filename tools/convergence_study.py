@@ -16,7 +16,7 @@ import torch.multiprocessing as mp
 
 
 def worker(rank, world, port, mode, scheme, epochs, q, nodes=4000,
-           edges=48000, use_gpu=False):
+           edges=48000, use_gpu=False, seed=0):
     os.environ.update(MASTER_ADDR='127.0.0.1', MASTER_PORT=str(port),
                       RANK=str(rank), WORLD_SIZE=str(world))
     from adaqp_amd.comm import Communicator
@@ -28,12 +28,12 @@ def worker(rank, world, port, mode, scheme, epochs, q, nodes=4000,
     from adaqp_amd.graph import random_partitioned_graph, partition_all
     comm = Communicator(backend='gloo')
     try:
-        torch.manual_seed(11)
+        torch.manual_seed(11 + seed)
         dev = torch.device('cuda:0') if use_gpu else torch.device('cpu')
         if use_gpu:
             torch.cuda.set_device(dev)
             comm.device = dev
-        g = random_partitioned_graph(nodes, edges, 32, 8, world, seed=21,
+        g = random_partitioned_graph(nodes, edges, 32, 8, world, seed=21 + seed,
                                      cut_frac=0.3, teacher_labels=True)
         lg = partition_all(g, world)[rank]
         engine = GraphEngine(lg, RunMode(mode), DistGNNType.DistGCN,
@@ -42,7 +42,7 @@ def worker(rank, world, port, mode, scheme, epochs, q, nodes=4000,
                             init_bits=8)
         if engine.bit_type.name == 'QUANT':
             assigner.initial_assignment()
-        torch.manual_seed(33)
+        torch.manual_seed(33 + seed)
         model = DistGCN(32, 64, 8, num_layers=3, dropout=0.0).to(dev)
         comm.sync_model_params(model)
         opt = torch.optim.Adam(model.parameters(), lr=0.01)
@@ -62,12 +62,12 @@ def worker(rank, world, port, mode, scheme, epochs, q, nodes=4000,
         Communicator.shutdown()
 
 
-def run(mode, scheme, epochs, port, nodes=4000, edges=48000, use_gpu=False):
+def run(mode, scheme, epochs, port, nodes=4000, edges=48000, use_gpu=False, seed=0):
     ctx = mp.get_context('spawn')
     q = ctx.SimpleQueue()
     procs = [ctx.Process(target=worker, args=(r, 2, port, mode, scheme,
                                               epochs, q, nodes, edges,
-                                              use_gpu))
+                                              use_gpu, seed))
              for r in range(2)]
     for p in procs:
         p.start()
@@ -82,6 +82,7 @@ def main():
     ap.add_argument('--epochs', type=int, default=200)
     ap.add_argument('--nodes', type=int, default=4000)
     ap.add_argument('--edges', type=int, default=48000)
+    ap.add_argument('--seed', type=int, default=0)
     ap.add_argument('--gpu', action='store_true',
                     help='2 ranks sharing cuda:0 (gloo-staged transport) — '
                          'runs the FULL adaptive pipeline on the HIP '
@@ -92,7 +93,7 @@ def main():
                                         ('AdaQP', 'adaptive'),
                                         ('AdaQP-q', 'uniform')]):
         m, s, curve = run(mode, scheme, args.epochs, 29720 + i,
-                          args.nodes, args.edges, args.gpu)
+                          args.nodes, args.edges, args.gpu, args.seed)
         results[f'{m}/{s}'] = curve
         print(f'{m}/{s}: final test acc {curve[-1]:.4f} '
               f'(best {max(curve):.4f})')
