@@ -43,7 +43,11 @@ class Communicator:
         self.rank = dist.get_rank()
         self.world_size = dist.get_world_size()
         if use_gpu:
-            local = int(os.environ.get('LOCAL_RANK', self.rank % max(torch.cuda.device_count(), 1)))
+            local = int(os.environ.get('LOCAL_RANK', self.rank))
+            # fold onto the available devices: on a full node this is the
+            # identity; on a smaller box several ranks share a GPU (the
+            # gloo-staged debug transport covers that case)
+            local %= max(torch.cuda.device_count(), 1)
             self.device = torch.device(f'cuda:{local}')
             torch.cuda.set_device(self.device)
         else:
