@@ -81,3 +81,28 @@ def test_bfs_alias_and_partition_all_grow():
     for lg in parts:
         lg.validate()
     assert sum(p.num_inner for p in parts) == gs.num_nodes
+
+
+def test_grow_unclustered_random_graph():
+    """No planted structure at all: grow must still terminate quickly,
+    assign everything, and stay balanced (cut quality is inherently
+    poor on a random graph — only the contract matters here)."""
+    import torch as t
+    from adaqp_amd.graph import GlobalGraph, grow_assignment
+    from adaqp_amd.graph.synthetic import _dedup_edges
+    gen = t.Generator().manual_seed(3)
+    n, m = 5000, 60000
+    s = t.randint(0, n, (m,), generator=gen)
+    d = t.randint(0, n, (m,), generator=gen)
+    s2 = t.cat([s, d, t.arange(n)])
+    d2 = t.cat([d, s, t.arange(n)])
+    s2, d2 = _dedup_edges(s2, d2, n)
+    g = GlobalGraph(n, s2, d2, t.randn(n, 4, generator=gen),
+                    t.randint(0, 3, (n,), generator=gen),
+                    t.ones(n, dtype=t.bool), t.zeros(n, dtype=t.bool),
+                    t.zeros(n, dtype=t.bool), 3, False)
+    ga = grow_assignment(g, 4, seed=0)
+    assert int((ga < 0).sum()) == 0
+    sizes = t.bincount(ga, minlength=4)
+    assert int(sizes.sum()) == n
+    assert float(sizes.max()) <= (n / 4) * 1.08 + 4
