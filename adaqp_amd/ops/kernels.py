@@ -107,8 +107,17 @@ def mixed_dequantize(payload: Tensor, params: Tensor, plan: SidePlan,
 _csr_cache: dict = {}
 
 import os as _os
-SEG_EDGES = int(_os.environ.get('ADAQP_SEG_EDGES', '256'))
-# max edges one sub-wavefront processes serially (env-tunable for A/B)
+# max edges one sub-wavefront processes serially. Auto default picks by
+# average degree (measured: 256 wins on low-degree products, 512 on
+# high-degree reddit — profiles/r02_NOTES.md); ADAQP_SEG_EDGES overrides.
+SEG_EDGES = int(_os.environ.get('ADAQP_SEG_EDGES', '0'))
+
+
+def _auto_seg(num_edges: int, num_rows: int) -> int:
+    if SEG_EDGES:
+        return SEG_EDGES
+    avg_deg = num_edges / max(num_rows, 1)
+    return 512 if avg_deg >= 256 else 256
 
 
 class SpmmView:
@@ -134,6 +143,7 @@ class SpmmView:
         self.base = int(base)
         self.nrows = int(nrows)
         self.col_blocked = col_block > 0 and indices.numel() > 0
+        SEG = _auto_seg(int(indices.numel()), nrows)
         indptr_c = indptr.cpu()
         counts = (indptr_c[1:] - indptr_c[:-1])
         if col_block > 0 and indices.numel():
@@ -152,13 +162,13 @@ class SpmmView:
             run_end = torch.cat([run_start[1:],
                                  torch.tensor([key.numel()])])
             run_len = run_end - run_start
-            nseg = (run_len + SEG_EDGES - 1) // SEG_EDGES
+            nseg = (run_len + SEG - 1) // SEG
             seg_runs = torch.repeat_interleave(
                 torch.arange(run_start.numel()), nseg)
             first = torch.repeat_interleave(torch.cumsum(nseg, 0) - nseg, nseg)
             in_run = torch.arange(seg_runs.numel(), dtype=torch.int64) - first
-            e0 = run_start[seg_runs] + in_run * SEG_EDGES
-            e1 = torch.minimum(e0 + SEG_EDGES, run_end[seg_runs])
+            e0 = run_start[seg_runs] + in_run * SEG
+            e1 = torch.minimum(e0 + SEG, run_end[seg_runs])
             seg_row = (key[run_start[seg_runs]] % nrows)
             segs_per_row = torch.bincount(seg_row, minlength=nrows)
             multi_mask = segs_per_row != 1     # incl. empty rows -> zeroed
@@ -169,12 +179,12 @@ class SpmmView:
             self.zero_rows = torch.nonzero(multi_mask, as_tuple=True)[0].to(torch.int32)
         else:
             self.indices = indices
-            nseg = torch.clamp((counts + SEG_EDGES - 1) // SEG_EDGES, min=1)
+            nseg = torch.clamp((counts + SEG - 1) // SEG, min=1)
             seg_row = torch.repeat_interleave(torch.arange(nrows, dtype=torch.int64), nseg)
             first = torch.repeat_interleave(torch.cumsum(nseg, 0) - nseg, nseg)
             seg_in_row = torch.arange(seg_row.numel(), dtype=torch.int64) - first
-            e0 = indptr_c[seg_row] + seg_in_row * SEG_EDGES
-            e1 = torch.minimum(e0 + SEG_EDGES, indptr_c[seg_row + 1])
+            e0 = indptr_c[seg_row] + seg_in_row * SEG
+            e1 = torch.minimum(e0 + SEG, indptr_c[seg_row + 1])
             multi_mask = nseg > 1
             self.seg_row = seg_row.to(torch.int32)
             self.seg_e0 = e0
