@@ -107,17 +107,15 @@ def mixed_dequantize(payload: Tensor, params: Tensor, plan: SidePlan,
 _csr_cache: dict = {}
 
 import os as _os
-# max edges one sub-wavefront processes serially. Auto default picks by
-# average degree (measured: 256 wins on low-degree products, 512 on
-# high-degree reddit — profiles/r02_NOTES.md); ADAQP_SEG_EDGES overrides.
-SEG_EDGES = int(_os.environ.get('ADAQP_SEG_EDGES', '0'))
+# max edges one sub-wavefront processes serially. 256 measured best on
+# the flagship shape in a same-box sweep (128/192/256 flat, 512/1024
+# slower); cross-box deltas on other datasets are within run-to-run
+# noise (profiles/r02_NOTES.md). ADAQP_SEG_EDGES overrides.
+SEG_EDGES = int(_os.environ.get('ADAQP_SEG_EDGES', '256'))
 
 
 def _auto_seg(num_edges: int, num_rows: int) -> int:
-    if SEG_EDGES:
-        return SEG_EDGES
-    avg_deg = num_edges / max(num_rows, 1)
-    return 512 if avg_deg >= 256 else 256
+    return SEG_EDGES
 
 
 class SpmmView:
