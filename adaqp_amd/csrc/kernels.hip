@@ -80,6 +80,8 @@ __global__ void quant_pack_kernel(
     int64_t n, int64_t F, int64_t ld, uint32_t seed,
     uint8_t* __restrict__ payload, uint16_t* __restrict__ params) {
     constexpr int VPB = 8 / BITS;          // values per byte
+    constexpr int MAXV = 16;               // staged floats per lane
+    constexpr int MAXCH = MAXV / VPB;      // register path covers F <= WAVE*MAXV
     const int64_t wid = (int64_t)blockIdx.x * (blockDim.x / WAVE)
                       + threadIdx.x / WAVE;
     if (wid >= n) return;
@@ -88,15 +90,37 @@ __global__ void quant_pack_kernel(
     const T* xr = x + row * ld;
     const int64_t bpn = (F * BITS + 7) / 8;
 
-    // pass 1: min/max (coalesced vpb-wide vector loads)
+    // single pass over x: stage this lane's values in registers while
+    // reducing min/max (the round-1 kernel read x twice). F beyond the
+    // register budget falls back to the two-pass read below.
+    const bool staged = F <= (int64_t)WAVE * MAXV;
+    float vals[MAXV];
     float mn = 1e38f, mx = -1e38f;
-    for (int64_t f0 = (int64_t)lane * VPB; f0 < F; f0 += (int64_t)WAVE * VPB) {
+    if (staged) {
 #pragma unroll
-        for (int k = 0; k < VPB; ++k) {
-            if (f0 + k < F) {
-                float v = to_f32<T>(xr[f0 + k]);
-                mn = fminf(mn, v);
-                mx = fmaxf(mx, v);
+        for (int c = 0; c < MAXCH; ++c) {
+            const int64_t f0 = ((int64_t)lane + (int64_t)c * WAVE) * VPB;
+            if (f0 >= F) break;
+#pragma unroll
+            for (int k = 0; k < VPB; ++k) {
+                float v = 0.f;
+                if (f0 + k < F) {
+                    v = to_f32<T>(xr[f0 + k]);
+                    mn = fminf(mn, v);
+                    mx = fmaxf(mx, v);
+                }
+                vals[c * VPB + k] = v;
+            }
+        }
+    } else {
+        for (int64_t f0 = (int64_t)lane * VPB; f0 < F; f0 += (int64_t)WAVE * VPB) {
+#pragma unroll
+            for (int k = 0; k < VPB; ++k) {
+                if (f0 + k < F) {
+                    float v = to_f32<T>(xr[f0 + k]);
+                    mn = fminf(mn, v);
+                    mx = fmaxf(mx, v);
+                }
             }
         }
     }
@@ -118,22 +142,45 @@ __global__ void quant_pack_kernel(
     }
     const uint32_t tag = (uint32_t)p;
 
-    // pass 2: quantize + pack one byte per lane-step
+    // quantize + pack one byte per lane-step. The staged path is fully
+    // unrolled so vals[] indices are compile-time constants (registers,
+    // no scratch).
     uint8_t* out = payload + off[wid];
-    for (int64_t b0 = lane; b0 * VPB < F; b0 += WAVE) {
-        uint32_t byte = 0;
+    if (staged) {
 #pragma unroll
-        for (int k = 0; k < VPB; ++k) {
-            const int64_t f = b0 * VPB + k;
-            if (f < F && scale > 0.f) {
-                float v = (to_f32<T>(xr[f]) - rmin) * scale;
-                float u = uniform01(seed, tag, (uint32_t)f);
-                int q = (int)floorf(v + u);
-                q = max(0, min(q, (1 << BITS) - 1));
-                byte |= ((uint32_t)q) << (k * BITS);
+        for (int c = 0; c < MAXCH; ++c) {
+            const int64_t b0 = (int64_t)lane + (int64_t)c * WAVE;
+            if (b0 * VPB >= F) break;
+            uint32_t byte = 0;
+#pragma unroll
+            for (int k = 0; k < VPB; ++k) {
+                const int64_t f = b0 * VPB + k;
+                if (f < F && scale > 0.f) {
+                    float v = (vals[c * VPB + k] - rmin) * scale;
+                    float u = uniform01(seed, tag, (uint32_t)f);
+                    int q = (int)floorf(v + u);
+                    q = max(0, min(q, (1 << BITS) - 1));
+                    byte |= ((uint32_t)q) << (k * BITS);
+                }
             }
+            out[b0] = (uint8_t)byte;
         }
-        out[b0] = (uint8_t)byte;
+    } else {
+        for (int64_t b0 = lane; b0 * VPB < F; b0 += WAVE) {
+            uint32_t byte = 0;
+#pragma unroll
+            for (int k = 0; k < VPB; ++k) {
+                const int64_t f = b0 * VPB + k;
+                if (f < F && scale > 0.f) {
+                    float v = (to_f32<T>(xr[f]) - rmin) * scale;
+                    float u = uniform01(seed, tag, (uint32_t)f);
+                    int q = (int)floorf(v + u);
+                    q = max(0, min(q, (1 << BITS) - 1));
+                    byte |= ((uint32_t)q) << (k * BITS);
+                }
+            }
+            out[b0] = (uint8_t)byte;
+        }
     }
 }
 
