@@ -6,8 +6,10 @@
 // AdaQP/model/ops.py:30). MI355X-first redesign:
 //
 //  * quant_pack fuses per-node min/max + scale + stochastic round + bit-pack
-//    + the per-(peer,bit-group) wire scatter into ONE pass (the reference
-//    computes rmin/rmax with two torch reductions in Python first).
+//    + the per-(peer,bit-group) wire scatter into ONE kernel and ONE read
+//    of x (lane-register staging for F<=1024; the reference computes
+//    rmin/rmax with two torch reductions in Python first, then re-reads
+//    x in the pack kernel).
 //  * packing is along the FEATURE axis so each node is a contiguous byte
 //    run: lane l owns (8/bits) consecutive features and emits whole bytes —
 //    coalesced float4/float2/float loads and byte stores across the
@@ -17,7 +19,8 @@
 //  * spmm_csr: one SUB-wavefront per destination-row segment (hub rows are
 //    split and atomically combined), dtype-templated (fp32 / bf16 with fp32
 //    accumulate), vector width chosen so row-base loads stay aligned, fused
-//    src/dst degree scaling, XCD-aware block swizzle.
+//    src/dst degree scaling, XCD-aware block swizzle, int32 column indices
+//    + segment bounds (half the index bytes), four gather loads in flight.
 //
 // Built for gfx950 only. No CUDA compatibility path.
 #include <hip/hip_runtime.h>
