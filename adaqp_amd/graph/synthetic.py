@@ -84,10 +84,12 @@ def synth_graph(name: str, num_parts: int, seed: int = 0, cut_frac: float = 0.10
 
 
 def _teacher_labels(src: Tensor, dst: Tensor, feats: Tensor, num_classes: int,
-                    gen: torch.Generator) -> Tensor:
+                    gen: torch.Generator, multilabel: bool = False) -> Tensor:
     """Labels from a random 1-hop mean-aggregation linear teacher, so synthetic
     graphs are LEARNABLE (accuracy comparisons Vanilla vs AdaQP are
-    meaningful — the reference uses real labeled datasets)."""
+    meaningful — the reference uses real labeled datasets). Multilabel:
+    each class is positive where its teacher logit is in the top decile
+    (~10% positive rate, yelp-like)."""
     n, f = feats.shape
     dev = 'cuda' if torch.cuda.is_available() else 'cpu'
     fd = feats.to(dev)
@@ -97,7 +99,14 @@ def _teacher_labels(src: Tensor, dst: Tensor, feats: Tensor, num_classes: int,
     agg.index_add_(0, dd, fd[sd])
     h = agg / deg[:, None]
     wout = torch.randn(f, num_classes, generator=gen).to(dev)
-    return (h @ wout).argmax(dim=1).cpu()
+    logits = h @ wout
+    if multilabel:
+        # per-class 90th percentile via sort (torch.quantile caps at
+        # 16M elements; amazonProducts logits are 168M)
+        k = min(int(0.9 * n), n - 1)
+        thr = logits.float().sort(dim=0).values[k:k + 1]
+        return (logits > thr).float().cpu()
+    return logits.argmax(dim=1).cpu()
 
 
 def random_partitioned_graph(num_nodes: int, num_edges: int, feat_dim: int,
@@ -133,7 +142,9 @@ def random_partitioned_graph(num_nodes: int, num_edges: int, feat_dim: int,
     s, d = _dedup_edges(s[keep], d[keep], num_nodes)
 
     feats = torch.randn(num_nodes, feat_dim, generator=g)
-    if multilabel:
+    if multilabel and teacher_labels:
+        labels = _teacher_labels(s, d, feats, num_classes, g, multilabel=True)
+    elif multilabel:
         labels = (torch.rand(num_nodes, num_classes, generator=g) < 0.1).float()
     elif teacher_labels:
         labels = _teacher_labels(s, d, feats, num_classes, g)
