@@ -52,6 +52,11 @@ def main():
 
     # allow plain `python main.py` without torchrun (world_size 1)
     import os
+    if os.environ.get('ADAQP_HANG_DEBUG'):
+        # dump all thread stacks and exit if a run wedges for N seconds
+        import faulthandler
+        faulthandler.dump_traceback_later(
+            int(os.environ['ADAQP_HANG_DEBUG']), exit=True)
     os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
     os.environ.setdefault('MASTER_PORT', '29501')
     os.environ.setdefault('RANK', '0')
