@@ -106,7 +106,35 @@ def grow_assignment(g: GlobalGraph, num_parts: int, seed: int = 0,
         assign = _grow_core(src, dst, n, P, node_w, None, seeds, gen, device)
 
     assign = _refine(assign, src, dst, n, P, imbalance, refine_passes)
+    assign = _force_balance(assign, src, dst, n, P, imbalance)
     return assign.cpu()
+
+
+def _force_balance(assign: Tensor, src: Tensor, dst: Tensor, n: int, P: int,
+                   imbalance: float) -> Tensor:
+    """Hard balance guarantee: gain-based refinement leaves parts over
+    quota when their nodes all prefer to stay (pathological sparse /
+    disconnected graphs inflate one part via quota bumps). Move the
+    least-internally-connected nodes out of oversized parts into the
+    emptiest ones until every part fits (n/P)*(1+imbalance)+1 — balance
+    is a memory contract per rank, cut quality is secondary here."""
+    hi = int((n / P) * (1 + imbalance)) + 1
+    sizes = torch.bincount(assign, minlength=P)
+    for _ in range(4 * P):
+        over = int(torch.argmax(sizes))
+        if int(sizes[over]) <= hi:
+            break
+        counts = torch.bincount(dst * P + assign[src],
+                                minlength=n * P).view(n, P)
+        cur = counts.gather(1, assign[:, None]).squeeze(1)
+        tgt = int(torch.argmin(sizes))
+        room = int(hi - sizes[tgt])
+        need = min(int(sizes[over] - hi), max(room, 1))
+        members = torch.nonzero(assign == over, as_tuple=True)[0]
+        weakest = members[torch.argsort(cur[members])[:need]]
+        assign[weakest] = tgt
+        sizes = torch.bincount(assign, minlength=P)
+    return assign
 
 
 def _label_prop_clusters(src: Tensor, dst: Tensor, n: int, cap: int,
