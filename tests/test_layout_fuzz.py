@@ -79,3 +79,34 @@ def test_grow_contract_fuzz(n, m, p, seed):
     assert int(sizes.sum()) == n
     # balance within the documented tolerance (+ slack for tiny parts)
     assert float(sizes.max()) <= (n / p) * 1.10 + p + 1
+
+
+@settings(max_examples=15, deadline=None)
+@given(
+    n=st.integers(30, 200),
+    parts=st.integers(2, 4),
+    seed=st.integers(0, 10**6),
+    cut=st.floats(0.05, 0.6),
+)
+def test_exchange_structure_fuzz(n, parts, seed, cut):
+    """For random partitioned graphs: every rank's LocalGraph validates,
+    and the send/recv structure is globally consistent — rank r sends to
+    q exactly the global ids q stores as r's remote block, in order."""
+    from adaqp_amd.graph import random_partitioned_graph, partition_all
+    g = random_partitioned_graph(n, 6 * n, 4, 3, parts, seed=seed,
+                                 cut_frac=cut)
+    lgs = partition_all(g, parts)
+    assert sum(p.num_inner for p in lgs) == n
+    for r, lg in enumerate(lgs):
+        lg.validate()
+        for q, other in enumerate(lgs):
+            if q == r:
+                continue
+            # ids r sends to q (global, in r's send order)
+            send_ids = (lg.local_to_global[lg.send_idx[q]]
+                        if q in lg.send_idx else torch.empty(0, dtype=torch.int64))
+            # q's remote block slice owned by r (global ids, stored order)
+            base = other.num_inner + sum(other.recv_splits[:r])
+            cnt = other.recv_splits[r]
+            recv_ids = other.local_to_global[base:base + cnt]
+            assert torch.equal(send_ids, recv_ids), (r, q)
