@@ -51,7 +51,7 @@ class Trainer:
         self.args = args
         self.cfg = load_config(args.dataset)
         rt = self.cfg['runtime']
-        for k in ('num_epochs', 'lr', 'log_steps'):
+        for k in ('num_epochs', 'lr', 'log_steps', 'eval_every'):
             v = getattr(args, k, None)
             if v is not None:
                 rt[k] = v

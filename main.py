@@ -40,6 +40,8 @@ def main():
     p.add_argument('--num_epochs', type=int, default=None)
     p.add_argument('--lr', type=float, default=None)
     p.add_argument('--log_steps', type=int, default=None)
+    p.add_argument('--eval_every', type=int, default=None,
+                   help='evaluate every N epochs (default: config, 1)')
     p.add_argument('--seed', type=int, default=None)
     p.add_argument('--dtype', type=str, default='fp32', choices=['fp32', 'bf16'])
     p.add_argument('--ckpt_path', type=str, default=None)
