@@ -4,9 +4,9 @@
 set -e
 SCALE=${1:-0.3}
 OUT=${2:-profiles/capture}
-mkdir -p "$OUT"
+REPO=$(cd "$(dirname "$0")/.." && pwd)     # resolve BEFORE leaving cwd
+mkdir -p "$REPO/$OUT"
 cd /tmp && export TMPDIR=/tmp
-REPO=$(cd "$(dirname "$0")/.." && pwd)
 # kernel trace + per-kernel stats (PMC counters crash rocprofv3 on this pool;
 # see profiles/r01_NOTES.md)
 rocprofv3 --kernel-trace --stats --output-format csv -d "$REPO/$OUT" -o bench \
