@@ -1,5 +1,6 @@
 #!/bin/bash
 # Reddit 4-part GCN, all four modes (reference: scripts/example/reddit_*.sh)
+cd "$(dirname "$0")/.."
 set -e
 for MODE in Vanilla AdaQP-q AdaQP-p AdaQP; do
   python -m torch.distributed.run --nnodes=1 --nproc-per-node 4 \

@@ -1,5 +1,6 @@
 #!/bin/bash
 # Yelp 4-part GraphSAGE (multilabel micro-F1), Vanilla vs AdaQP
+cd "$(dirname "$0")/.."
 set -e
 for MODE in Vanilla AdaQP; do
   python -m torch.distributed.run --nnodes=1 --nproc-per-node 4 \
